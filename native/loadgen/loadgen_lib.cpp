@@ -1,0 +1,281 @@
+// loadgen_lib.cpp — host-side C API around the CDNA4 load kernels.
+//
+// MI355X-native replacement for the reference's load workload machinery
+// (reference: cuda-test-deployment.yaml:18-19 runs 5000 sequential CUDA
+// `vectorAdd` *processes*; README.md:115 doubles the load with a second
+// loop). Here the same load shapes are produced from one resident process:
+//   * lg_vector_add_loop(): N launches of the vectorAdd kernel with a host
+//     sync between launches — reproduces the reference's partial-utilization
+//     shape (launch/driver overhead dominates) without 5000 process forks.
+//   * lg_gemm_burn(): duty-cycled MFMA bf16 GEMM bursts targeting a given
+//     GPU-busy percentage — the high/tunable load the scale-up experiments
+//     need (reference has no equivalent).
+//   * lg_*_verify(): run one kernel on caller-provided host data so tests
+//     can check numerics against a plain fp32 reference.
+//
+// Exposed as a plain C ABI: consumed by loadgen_main.cpp (CLI) and by
+// mi355x_gpu_hpa/loadgen (ctypes).
+
+#include <hip/hip_runtime.h>
+
+#include <chrono>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <thread>
+#include <vector>
+
+extern "C" __global__ void vector_add_f32(const float*, const float*, float*, int);
+extern "C" __global__ void vector_add_f32x4(const float4*, const float4*, float4*, int);
+extern "C" __global__ void gemm_bf16_tn(const unsigned short*, const unsigned short*,
+                                        float*, int, int, int, int);
+
+#define LG_CHECK(expr)                                                        \
+    do {                                                                      \
+        hipError_t _e = (expr);                                               \
+        if (_e != hipSuccess) {                                               \
+            std::snprintf(g_last_error, sizeof(g_last_error), "%s:%d %s: %s", \
+                          __FILE__, __LINE__, #expr, hipGetErrorString(_e));  \
+            return -1;                                                        \
+        }                                                                     \
+    } while (0)
+
+static char g_last_error[512] = "";
+
+static inline double now_ms()
+{
+    return std::chrono::duration<double, std::milli>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+}
+
+static inline unsigned short f32_to_bf16(float f)
+{
+    uint32_t u;
+    std::memcpy(&u, &f, 4);
+    // round-to-nearest-even
+    uint32_t lsb = (u >> 16) & 1;
+    u += 0x7fffu + lsb;
+    return (unsigned short)(u >> 16);
+}
+
+extern "C" {
+
+const char* lg_last_error() { return g_last_error; }
+
+int lg_device_count()
+{
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}
+
+// ---------------------------------------------------------------------------
+// vectorAdd
+// ---------------------------------------------------------------------------
+
+// One allocation + `iters` kernel launches (sync per launch, like one
+// process-per-launch in the reference but without exec overhead).
+// Returns 0 on success; *ms_out = total wall ms over the launches.
+int lg_vector_add_loop(int device, int n, int iters, double* ms_out)
+{
+    LG_CHECK(hipSetDevice(device));
+    float *a, *b, *c;
+    size_t bytes = (size_t)n * 4;
+    LG_CHECK(hipMalloc(&a, bytes));
+    LG_CHECK(hipMalloc(&b, bytes));
+    LG_CHECK(hipMalloc(&c, bytes));
+    std::vector<float> h(n);
+    for (int i = 0; i < n; ++i) h[i] = (float)((i * 2654435761u) % 1000) * 1e-3f;
+    LG_CHECK(hipMemcpy(a, h.data(), bytes, hipMemcpyHostToDevice));
+    LG_CHECK(hipMemcpy(b, h.data(), bytes, hipMemcpyHostToDevice));
+
+    int threads = 256;
+    int blocks = (n + threads - 1) / threads;
+    if (blocks > 4096) blocks = 4096; // grid-stride covers the rest
+    double t0 = now_ms();
+    for (int i = 0; i < iters; ++i) {
+        hipLaunchKernelGGL(vector_add_f32, dim3(blocks), dim3(threads), 0, 0, a, b, c, n);
+        LG_CHECK(hipDeviceSynchronize());
+    }
+    double t1 = now_ms();
+    if (ms_out) *ms_out = t1 - t0;
+    LG_CHECK(hipFree(a));
+    LG_CHECK(hipFree(b));
+    LG_CHECK(hipFree(c));
+    return 0;
+}
+
+// Numerics entry: c_out[i] = a[i] + b[i] computed on the GPU.
+int lg_vector_add_verify(int device, const float* a_h, const float* b_h, float* c_out, int n)
+{
+    LG_CHECK(hipSetDevice(device));
+    float *a, *b, *c;
+    size_t bytes = (size_t)n * 4;
+    LG_CHECK(hipMalloc(&a, bytes));
+    LG_CHECK(hipMalloc(&b, bytes));
+    LG_CHECK(hipMalloc(&c, bytes));
+    LG_CHECK(hipMemcpy(a, a_h, bytes, hipMemcpyHostToDevice));
+    LG_CHECK(hipMemcpy(b, b_h, bytes, hipMemcpyHostToDevice));
+    int threads = 256;
+    if (n % 4 == 0) {
+        int n4 = n / 4;
+        int blocks = (n4 + threads - 1) / threads;
+        if (blocks > 4096) blocks = 4096;
+        hipLaunchKernelGGL(vector_add_f32x4, dim3(blocks), dim3(threads), 0, 0,
+                           (const float4*)a, (const float4*)b, (float4*)c, n4);
+    } else {
+        int blocks = (n + threads - 1) / threads;
+        if (blocks > 4096) blocks = 4096;
+        hipLaunchKernelGGL(vector_add_f32, dim3(blocks), dim3(threads), 0, 0, a, b, c, n);
+    }
+    LG_CHECK(hipDeviceSynchronize());
+    LG_CHECK(hipMemcpy(c_out, c, bytes, hipMemcpyDeviceToHost));
+    LG_CHECK(hipFree(a));
+    LG_CHECK(hipFree(b));
+    LG_CHECK(hipFree(c));
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// MFMA bf16 GEMM
+// ---------------------------------------------------------------------------
+
+struct GemmBufs {
+    unsigned short* a = nullptr;
+    unsigned short* bt = nullptr;
+    float* c = nullptr;
+    int m = 0, n = 0, k = 0;
+};
+
+static int gemm_alloc(GemmBufs& g, int m, int n, int k, bool fill_random)
+{
+    LG_CHECK(hipMalloc(&g.a, (size_t)m * k * 2));
+    LG_CHECK(hipMalloc(&g.bt, (size_t)n * k * 2));
+    LG_CHECK(hipMalloc(&g.c, (size_t)m * n * 4));
+    g.m = m; g.n = n; g.k = k;
+    if (fill_random) {
+        // Random-ish bf16 in [-1, 1): DVFS-honest load (zero-filled operands
+        // clock higher and overstate TF/s — playbook §5.4 rule 25).
+        size_t na = (size_t)m * k, nb = (size_t)n * k;
+        std::vector<unsigned short> h(na > nb ? na : nb);
+        uint32_t s = 0x12345678u;
+        for (size_t i = 0; i < h.size(); ++i) {
+            s = s * 1664525u + 1013904223u;
+            float f = ((s >> 8) & 0xffff) / 32768.0f - 1.0f;
+            h[i] = f32_to_bf16(f);
+        }
+        LG_CHECK(hipMemcpy(g.a, h.data(), na * 2, hipMemcpyHostToDevice));
+        LG_CHECK(hipMemcpy(g.bt, h.data(), nb * 2, hipMemcpyHostToDevice));
+    }
+    return 0;
+}
+
+static void gemm_free(GemmBufs& g)
+{
+    (void)hipFree(g.a); (void)hipFree(g.bt); (void)hipFree(g.c);
+    g = GemmBufs{};
+}
+
+static int gemm_launch(const GemmBufs& g, hipStream_t stream)
+{
+    int n_tiles = (g.m / 128) * (g.n / 128);
+    int blocks = n_tiles < 2048 ? n_tiles : 2048;
+    int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
+    hipLaunchKernelGGL(gemm_bf16_tn, dim3(blocks), dim3(256), 0, stream,
+                       g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
+    return 0;
+}
+
+// Timed GEMM: `iters` back-to-back launches after `warmup` untimed ones.
+// *ms_out = mean ms per GEMM, *tflops_out = 2*M*N*K / time.
+int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
+                       double* ms_out, double* tflops_out)
+{
+    if (m % 128 || n % 128 || k % 64) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "gemm dims must be multiples of 128/128/64");
+        return -1;
+    }
+    LG_CHECK(hipSetDevice(device));
+    GemmBufs g;
+    if (gemm_alloc(g, m, n, k, true)) return -1;
+    for (int i = 0; i < warmup; ++i) gemm_launch(g, 0);
+    LG_CHECK(hipDeviceSynchronize());
+    double t0 = now_ms();
+    for (int i = 0; i < iters; ++i) gemm_launch(g, 0);
+    LG_CHECK(hipDeviceSynchronize());
+    double t1 = now_ms();
+    double ms = (t1 - t0) / iters;
+    if (ms_out) *ms_out = ms;
+    if (tflops_out) *tflops_out = 2.0 * m * n * k / (ms * 1e-3) / 1e12;
+    gemm_free(g);
+    return 0;
+}
+
+// Numerics entry: C f32 = A bf16 @ B^T bf16 on caller data (A row-major
+// [m][k] as f32 -> converted; Bt row-major [n][k]).
+int lg_gemm_bf16_verify(int device, const float* a_h, const float* bt_h,
+                        float* c_out, int m, int n, int k)
+{
+    if (m % 128 || n % 128 || k % 64) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "gemm dims must be multiples of 128/128/64");
+        return -1;
+    }
+    LG_CHECK(hipSetDevice(device));
+    GemmBufs g;
+    if (gemm_alloc(g, m, n, k, false)) return -1;
+    std::vector<unsigned short> tmp((size_t)m * k);
+    for (size_t i = 0; i < tmp.size(); ++i) tmp[i] = f32_to_bf16(a_h[i]);
+    LG_CHECK(hipMemcpy(g.a, tmp.data(), tmp.size() * 2, hipMemcpyHostToDevice));
+    tmp.resize((size_t)n * k);
+    for (size_t i = 0; i < tmp.size(); ++i) tmp[i] = f32_to_bf16(bt_h[i]);
+    LG_CHECK(hipMemcpy(g.bt, tmp.data(), tmp.size() * 2, hipMemcpyHostToDevice));
+    gemm_launch(g, 0);
+    LG_CHECK(hipDeviceSynchronize());
+    LG_CHECK(hipGetLastError());
+    LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
+    gemm_free(g);
+    return 0;
+}
+
+// Duty-cycled GEMM burn: aim at `target_util_pct` GPU-busy for `seconds`.
+// Open-loop duty cycle over a `period_ms` window: run GEMM launches for
+// util*period, sleep the rest. The exporter (or rocm-smi) measures the
+// resulting busy%; tests close the loop.
+// stop_flag: optional; polled between periods (set non-zero to stop early).
+int lg_gemm_burn(int device, double target_util_pct, double seconds,
+                 int m, int n, int k, double period_ms, volatile int* stop_flag)
+{
+    if (m <= 0) m = 4096;
+    if (n <= 0) n = 4096;
+    if (k <= 0) k = 4096;
+    if (period_ms <= 0) period_ms = 100.0;
+    if (target_util_pct < 0) target_util_pct = 0;
+    if (target_util_pct > 100) target_util_pct = 100;
+    LG_CHECK(hipSetDevice(device));
+    GemmBufs g;
+    if (gemm_alloc(g, m, n, k, true)) return -1;
+    // one calibration launch so the first period isn't all compile/warmup
+    gemm_launch(g, 0);
+    LG_CHECK(hipDeviceSynchronize());
+
+    double t_end = now_ms() + seconds * 1e3;
+    while (now_ms() < t_end) {
+        if (stop_flag && *stop_flag) break;
+        double period_start = now_ms();
+        double busy_until = period_start + period_ms * target_util_pct / 100.0;
+        while (now_ms() < busy_until) {
+            gemm_launch(g, 0);
+            LG_CHECK(hipDeviceSynchronize());
+        }
+        double rest = period_start + period_ms - now_ms();
+        if (rest > 0)
+            std::this_thread::sleep_for(std::chrono::duration<double, std::milli>(rest));
+    }
+    gemm_free(g);
+    return 0;
+}
+
+} // extern "C"
