@@ -101,7 +101,7 @@ extern "C" __global__ void __launch_bounds__(256, 1) gemm_bf16_tn(
             const unsigned short* ga = A + (row0 + ld_row) * (long)K + ld_k;
             const unsigned short* gb = Bt + (col0 + ld_row) * (long)K + ld_k;
             unsigned short* la = &lds[0];
-            unsigned short* lb = &lds[2 * BM * BK];
+            unsigned short* lb = &lds[BM * BK];
 #pragma unroll
             for (int it = 0; it < GLDS_PER_TILE_PER_WAVE; ++it) {
                 __builtin_amdgcn_global_load_lds(
