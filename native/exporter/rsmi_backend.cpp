@@ -1,0 +1,259 @@
+// rsmi_backend.cpp — gfx950 counter sampling via rocm_smi_lib.
+//
+// Replaces the reference's DCGM+NVML layer (host libs mounted
+// at dcgm-exporter.yaml:53-62, SYS_ADMIN at :42-48). No DCGM, no NVML: we
+// talk to the amdgpu KFD interfaces through librocm_smi64, which needs only
+// /dev/kfd + /dev/dri access — no privileged container.
+//
+// librocm_smi64 is resolved with dlopen so the exporter binary starts (and
+// can serve its mock backend) on boxes without ROCm installed; on a GPU node
+// it must resolve, and every sampling failure is surfaced per-metric rather
+// than crashing the daemon.
+
+#include "backend.h"
+
+#include <dlfcn.h>
+#include <rocm_smi/rocm_smi.h>
+
+#include <chrono>
+#include <cstdio>
+#include <cstring>
+
+namespace mi355x {
+
+namespace {
+
+double now_ms()
+{
+    return std::chrono::duration<double, std::milli>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+}
+
+struct RsmiApi {
+    void* handle = nullptr;
+#define DECL(fn) decltype(&::fn) fn = nullptr
+    DECL(rsmi_init);
+    DECL(rsmi_shut_down);
+    DECL(rsmi_num_monitor_devices);
+    DECL(rsmi_dev_name_get);
+    DECL(rsmi_dev_unique_id_get);
+    DECL(rsmi_dev_pci_id_get);
+    DECL(rsmi_dev_drm_render_minor_get);
+    DECL(rsmi_dev_busy_percent_get);
+    DECL(rsmi_dev_memory_busy_percent_get);
+    DECL(rsmi_dev_memory_usage_get);
+    DECL(rsmi_dev_memory_total_get);
+    DECL(rsmi_dev_temp_metric_get);
+    DECL(rsmi_dev_power_ave_get);
+    DECL(rsmi_dev_energy_count_get);
+    DECL(rsmi_dev_gpu_clk_freq_get);
+    DECL(rsmi_dev_gpu_metrics_info_get);
+#undef DECL
+
+    bool load(std::string* err)
+    {
+        const char* names[] = {"librocm_smi64.so.1", "librocm_smi64.so",
+                               "/opt/rocm/lib/librocm_smi64.so"};
+        for (const char* n : names) {
+            handle = dlopen(n, RTLD_NOW | RTLD_GLOBAL);
+            if (handle) break;
+        }
+        if (!handle) {
+            if (err) *err = std::string("dlopen librocm_smi64 failed: ") + dlerror();
+            return false;
+        }
+#define RESOLVE(fn)                                                     \
+    fn = reinterpret_cast<decltype(&::fn)>(dlsym(handle, #fn));         \
+    if (!fn) {                                                          \
+        if (err) *err = "dlsym " #fn " failed";                         \
+        return false;                                                   \
+    }
+        RESOLVE(rsmi_init)
+        RESOLVE(rsmi_shut_down)
+        RESOLVE(rsmi_num_monitor_devices)
+        RESOLVE(rsmi_dev_name_get)
+        RESOLVE(rsmi_dev_unique_id_get)
+        RESOLVE(rsmi_dev_pci_id_get)
+        RESOLVE(rsmi_dev_drm_render_minor_get)
+        RESOLVE(rsmi_dev_busy_percent_get)
+        RESOLVE(rsmi_dev_memory_busy_percent_get)
+        RESOLVE(rsmi_dev_memory_usage_get)
+        RESOLVE(rsmi_dev_memory_total_get)
+        RESOLVE(rsmi_dev_temp_metric_get)
+        RESOLVE(rsmi_dev_power_ave_get)
+        RESOLVE(rsmi_dev_energy_count_get)
+        RESOLVE(rsmi_dev_gpu_clk_freq_get)
+        RESOLVE(rsmi_dev_gpu_metrics_info_get)
+#undef RESOLVE
+        return true;
+    }
+};
+
+class RsmiBackend final : public Backend {
+  public:
+    RsmiBackend(RsmiApi api, uint32_t n) : api_(api), n_(n) {}
+    ~RsmiBackend() override { api_.rsmi_shut_down(); }
+
+    int device_count() override { return (int)n_; }
+    std::string name() const override { return "rocm_smi"; }
+
+    GpuInfo info(int dev) override
+    {
+        GpuInfo gi;
+        gi.index = dev;
+        char buf[256] = {0};
+        if (api_.rsmi_dev_name_get(dev, buf, sizeof(buf)) == RSMI_STATUS_SUCCESS)
+            gi.name = buf;
+        uint64_t uid = 0;
+        if (api_.rsmi_dev_unique_id_get(dev, &uid) == RSMI_STATUS_SUCCESS) {
+            char h[32];
+            std::snprintf(h, sizeof(h), "%016lx", (unsigned long)uid);
+            gi.uuid = h;
+        }
+        uint64_t bdf = 0;
+        if (api_.rsmi_dev_pci_id_get(dev, &bdf) == RSMI_STATUS_SUCCESS) {
+            // bdfid: [63:32] domain, [15:8] bus, [7:3] device, [2:0] function
+            char b[32];
+            std::snprintf(b, sizeof(b), "%04lx:%02x:%02x.%x",
+                          (unsigned long)(bdf >> 32) & 0xffffffff,
+                          (unsigned)(bdf >> 8) & 0xff, (unsigned)(bdf >> 3) & 0x1f,
+                          (unsigned)bdf & 0x7);
+            gi.pci_bdf = b;
+        }
+        uint32_t minor = 0;
+        if (api_.rsmi_dev_drm_render_minor_get(dev, &minor) == RSMI_STATUS_SUCCESS) {
+            gi.drm_render = "renderD" + std::to_string(minor);
+        }
+        return gi;
+    }
+
+    GpuSample sample(int dev) override
+    {
+        GpuSample s;
+        s.ts_ms = now_ms();
+
+        uint32_t pct = 0;
+        if (api_.rsmi_dev_busy_percent_get(dev, &pct) == RSMI_STATUS_SUCCESS) {
+            s.busy_pct = pct;
+            s.ok = true;
+        }
+        if (api_.rsmi_dev_memory_busy_percent_get(dev, &pct) == RSMI_STATUS_SUCCESS)
+            s.mem_busy_pct = pct;
+
+        uint64_t v = 0;
+        if (api_.rsmi_dev_memory_usage_get(dev, RSMI_MEM_TYPE_VRAM, &v) ==
+            RSMI_STATUS_SUCCESS)
+            s.vram_used_bytes = (double)v;
+        if (api_.rsmi_dev_memory_total_get(dev, RSMI_MEM_TYPE_VRAM, &v) ==
+            RSMI_STATUS_SUCCESS)
+            s.vram_total_bytes = (double)v;
+
+        int64_t t = 0;
+        if (api_.rsmi_dev_temp_metric_get(dev, RSMI_TEMP_TYPE_EDGE, RSMI_TEMP_CURRENT,
+                                          &t) == RSMI_STATUS_SUCCESS)
+            s.temp_edge_c = t / 1000.0;
+        if (api_.rsmi_dev_temp_metric_get(dev, RSMI_TEMP_TYPE_JUNCTION,
+                                          RSMI_TEMP_CURRENT, &t) == RSMI_STATUS_SUCCESS)
+            s.temp_hotspot_c = t / 1000.0;
+        if (api_.rsmi_dev_temp_metric_get(dev, RSMI_TEMP_TYPE_MEMORY, RSMI_TEMP_CURRENT,
+                                          &t) == RSMI_STATUS_SUCCESS)
+            s.temp_mem_c = t / 1000.0;
+
+        uint64_t uw = 0;
+        if (api_.rsmi_dev_power_ave_get(dev, 0, &uw) == RSMI_STATUS_SUCCESS)
+            s.power_w = uw / 1e6;
+
+        uint64_t energy = 0, ets = 0;
+        float res = 0;
+        if (api_.rsmi_dev_energy_count_get(dev, &energy, &res, &ets) ==
+            RSMI_STATUS_SUCCESS)
+            s.energy_j = energy * (double)res / 1e6; // counter*resolution = uJ
+
+        rsmi_frequencies_t f;
+        std::memset(&f, 0, sizeof(f));
+        if (api_.rsmi_dev_gpu_clk_freq_get(dev, RSMI_CLK_TYPE_SYS, &f) ==
+                RSMI_STATUS_SUCCESS &&
+            f.current < f.num_supported)
+            s.sclk_mhz = f.frequency[f.current] / 1e6;
+        std::memset(&f, 0, sizeof(f));
+        if (api_.rsmi_dev_gpu_clk_freq_get(dev, RSMI_CLK_TYPE_MEM, &f) ==
+                RSMI_STATUS_SUCCESS &&
+            f.current < f.num_supported)
+            s.mclk_mhz = f.frequency[f.current] / 1e6;
+
+        rsmi_gpu_metrics_t gm;
+        std::memset(&gm, 0, sizeof(gm));
+        if (api_.rsmi_dev_gpu_metrics_info_get(dev, &gm) == RSMI_STATUS_SUCCESS) {
+            auto u16ok = [](uint16_t x) { return x != 0xffff; };
+            if (u16ok(gm.average_gfx_activity)) {
+                s.gfx_activity_pct = gm.average_gfx_activity;
+                if (s.busy_pct < 0) s.busy_pct = gm.average_gfx_activity;
+                s.ok = true;
+            }
+            if (u16ok(gm.average_umc_activity)) {
+                s.umc_activity_pct = gm.average_umc_activity;
+                if (s.mem_busy_pct < 0) s.mem_busy_pct = gm.average_umc_activity;
+            }
+            if (gm.gfx_activity_acc != 0xffffffffu)
+                s.gfx_activity_acc = gm.gfx_activity_acc;
+            if (gm.mem_activity_acc != 0xffffffffu)
+                s.mem_activity_acc = gm.mem_activity_acc;
+            if (u16ok(gm.current_socket_power) && s.power_w < 0)
+                s.power_w = gm.current_socket_power;
+            if (u16ok(gm.current_gfxclk) && s.sclk_mhz < 0)
+                s.sclk_mhz = gm.current_gfxclk;
+            if (u16ok(gm.current_uclk) && s.mclk_mhz < 0)
+                s.mclk_mhz = gm.current_uclk;
+            if (u16ok(gm.temperature_mem) && s.temp_mem_c < 0)
+                s.temp_mem_c = gm.temperature_mem;
+
+            // PCIe: accumulated GB/s-seconds; expose instantaneous if present
+            if (gm.pcie_bandwidth_inst && gm.pcie_bandwidth_inst != ~0ull) {
+                s.pcie_tx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
+                s.pcie_rx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
+            }
+
+            if (u16ok(gm.xgmi_link_width)) s.xgmi_link_width = gm.xgmi_link_width;
+            if (u16ok(gm.xgmi_link_speed)) s.xgmi_link_speed_gbps = gm.xgmi_link_speed;
+            int nl = 0;
+            for (int i = 0; i < kMaxXgmiLinks; ++i) {
+                uint64_t r = gm.xgmi_read_data_acc[i];
+                uint64_t w = gm.xgmi_write_data_acc[i];
+                if (r == ~0ull) r = 0;
+                if (w == ~0ull) w = 0;
+                s.xgmi_read_acc_kb[i] = (double)r;
+                s.xgmi_write_acc_kb[i] = (double)w;
+                if (gm.xgmi_link_status[i] == 1 || r || w) nl = i + 1;
+            }
+            s.xgmi_num_links = nl;
+        }
+        return s;
+    }
+
+  private:
+    RsmiApi api_;
+    uint32_t n_;
+};
+
+} // namespace
+
+std::unique_ptr<Backend> make_rsmi_backend(std::string* err)
+{
+    RsmiApi api;
+    if (!api.load(err)) return nullptr;
+    rsmi_status_t st = api.rsmi_init(0);
+    if (st != RSMI_STATUS_SUCCESS) {
+        if (err) *err = "rsmi_init failed (status " + std::to_string((int)st) + ")";
+        return nullptr;
+    }
+    uint32_t n = 0;
+    if (api.rsmi_num_monitor_devices(&n) != RSMI_STATUS_SUCCESS || n == 0) {
+        api.rsmi_shut_down();
+        if (err) *err = "no AMD GPU devices visible to rocm_smi";
+        return nullptr;
+    }
+    return std::make_unique<RsmiBackend>(api, n);
+}
+
+} // namespace mi355x
