@@ -1,0 +1,19 @@
+from .hpa import HpaSpec, HpaState, desired_replicas, reconcile
+from .loop import (
+    REFERENCE_RULE_EXPR,
+    REFERENCE_RULE_NAME,
+    ControlLoop,
+    LoopResult,
+    RecordingRule,
+    synth_pod_labels,
+)
+from .promql import PromQLError, Sample, evaluate, evaluate_scalar
+from .scraper import Scraper, ScrapeTarget, parse_prometheus_text
+
+__all__ = [
+    "HpaSpec", "HpaState", "desired_replicas", "reconcile",
+    "ControlLoop", "LoopResult", "RecordingRule", "synth_pod_labels",
+    "REFERENCE_RULE_EXPR", "REFERENCE_RULE_NAME",
+    "PromQLError", "Sample", "evaluate", "evaluate_scalar",
+    "Scraper", "ScrapeTarget", "parse_prometheus_text",
+]

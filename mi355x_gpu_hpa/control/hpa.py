@@ -1,0 +1,80 @@
+"""hpa.py — the HorizontalPodAutoscaler reconciliation algorithm.
+
+Implements what the stock kube-controller-manager HPA loop does for an
+``Object``-type custom metric (the reference's cuda-test-hpa.yaml:13-21,
+``metricName: cuda_test_gpu_avg, targetValue: 5``; reconcile math per
+SURVEY.md §3.4):
+
+    desired = ceil(current / target * currentReplicas)
+    clamped to [minReplicas, maxReplicas]
+
+plus the controller's behavior details that matter for the scale-up curve:
+  * 10% tolerance band around ratio 1.0 (no scaling inside it)
+  * scale-down stabilization window (default 300 s: the highest desired
+    within the window wins, which is what delays downscale)
+  * missing metric => no change
+
+In a real cluster the stock controller does this (SURVEY.md C12 "reuse
+as-is"); this implementation drives the in-process control-loop harness and
+the latency bench, and is unit-tested against the documented edge cases.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+
+@dataclass
+class HpaSpec:
+    min_replicas: int = 1
+    max_replicas: int = 3          # reference default (cuda-test-hpa.yaml:11-12)
+    target_value: float = 5.0      # reference targetValue (cuda-test-hpa.yaml:21)
+    tolerance: float = 0.1         # upstream default
+    downscale_stabilization_s: float = 300.0  # upstream default
+
+
+@dataclass
+class HpaState:
+    current_replicas: int = 1
+    # (timestamp_s, desired) recommendations within the stabilization window
+    recommendations: List[Tuple[float, int]] = field(default_factory=list)
+
+
+def desired_replicas(
+    spec: HpaSpec, current_replicas: int, metric_value: Optional[float]
+) -> int:
+    """Raw desired-replica computation (no stabilization)."""
+    if metric_value is None:
+        return current_replicas
+    if spec.target_value <= 0:
+        raise ValueError("target_value must be positive")
+    ratio = metric_value / spec.target_value
+    if abs(ratio - 1.0) <= spec.tolerance:
+        return current_replicas
+    desired = math.ceil(ratio * current_replicas)
+    return max(spec.min_replicas, min(spec.max_replicas, desired))
+
+
+def reconcile(
+    spec: HpaSpec, state: HpaState, metric_value: Optional[float], now_s: float
+) -> int:
+    """One HPA sync: returns the new replica count (and updates state).
+
+    Scale-ups apply immediately; scale-downs are stabilized — the new count
+    is the MAX of desired values recommended within the stabilization
+    window, so a transient dip can't flap the deployment.
+    """
+    desired = desired_replicas(spec, state.current_replicas, metric_value)
+    state.recommendations.append((now_s, desired))
+    cutoff = now_s - spec.downscale_stabilization_s
+    state.recommendations = [(t, d) for (t, d) in state.recommendations if t >= cutoff]
+
+    if desired >= state.current_replicas:
+        new = desired
+    else:
+        new = max(d for (_, d) in state.recommendations)
+        new = max(new, spec.min_replicas)
+    state.current_replicas = new
+    return new

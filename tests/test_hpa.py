@@ -1,0 +1,73 @@
+"""Unit tests for the HPA reconcile algorithm (reference semantics:
+cuda-test-hpa.yaml:11-21, desired = ceil(current/target x replicas))."""
+
+from mi355x_gpu_hpa.control import HpaSpec, HpaState, desired_replicas, reconcile
+
+
+def spec(**kw):
+    defaults = dict(min_replicas=1, max_replicas=3, target_value=5.0)
+    defaults.update(kw)
+    return HpaSpec(**defaults)
+
+
+class TestDesired:
+    def test_within_tolerance_no_change(self):
+        assert desired_replicas(spec(), 2, 5.2) == 2  # ratio 1.04 < 1.1
+        assert desired_replicas(spec(), 2, 4.6) == 2  # ratio 0.92 > 0.9
+
+    def test_scale_up_ceil(self):
+        # the reference trigger: load ~8% vs target 5 on 1 replica
+        assert desired_replicas(spec(), 1, 8.0) == 2
+        assert desired_replicas(spec(), 1, 10.1) == 3
+
+    def test_clamp_max(self):
+        assert desired_replicas(spec(), 1, 100.0) == 3
+        assert desired_replicas(spec(max_replicas=8), 1, 100.0) == 8
+
+    def test_clamp_min(self):
+        assert desired_replicas(spec(), 3, 0.0) == 1
+
+    def test_missing_metric_no_change(self):
+        assert desired_replicas(spec(), 2, None) == 2
+
+    def test_scale_up_proportional(self):
+        # 2 replicas at 7.5 avg: ratio 1.5 -> ceil(3.0) = 3
+        assert desired_replicas(spec(), 2, 7.5) == 3
+
+
+class TestReconcile:
+    def test_scale_up_immediate(self):
+        st = HpaState(current_replicas=1)
+        assert reconcile(spec(), st, 20.0, now_s=0.0) == 3
+        assert st.current_replicas == 3
+
+    def test_scale_down_stabilized(self):
+        s = spec(downscale_stabilization_s=300)
+        st = HpaState(current_replicas=3)
+        # load drops: desired=1, but within the window the max recommendation
+        # (3, from the moment of the drop onwards only 1s) holds
+        reconcile(s, st, 20.0, now_s=0.0)       # records desired=3
+        assert reconcile(s, st, 0.0, now_s=10.0) == 3
+        assert reconcile(s, st, 0.0, now_s=200.0) == 3
+        # after the window expires, downscale lands
+        assert reconcile(s, st, 0.0, now_s=301.0) == 1
+
+    def test_flapping_suppressed(self):
+        s = spec(downscale_stabilization_s=60)
+        st = HpaState(current_replicas=2)
+        reconcile(s, st, 10.0, now_s=0.0)   # up to 3... wait ratio 2 on 2 -> 3 (clamped)
+        assert st.current_replicas == 3
+        assert reconcile(s, st, 4.9, now_s=10.0) == 3   # dip within window
+        assert reconcile(s, st, 10.0, now_s=20.0) == 3  # back up
+
+    def test_scale_up_curve_to_eight(self):
+        # config 4: 1 -> 8 replica scale-up under sustained high load
+        s = spec(max_replicas=8)
+        st = HpaState(current_replicas=1)
+        seen = [st.current_replicas]
+        for t in range(6):
+            # each replica still saturated at 40% (target 5) => keep scaling
+            reconcile(s, st, 40.0, now_s=float(t * 15))
+            seen.append(st.current_replicas)
+        assert st.current_replicas == 8
+        assert seen[0] == 1 and sorted(seen) == seen  # monotone ramp
