@@ -1,0 +1,249 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: the closed-loop GPU-metric autoscaling
+pipeline on MI355X, under real MFMA bf16 GEMM load.
+
+Measures BASELINE.json's headline metric: **p50 scrape->HPA-scale latency**
+— the time for one full control cycle (scrape the native exporter's
+/metrics for every GPU -> evaluate the reference recording rule
+(cuda-test-prometheusrule.yaml:13 semantics) -> HPA reconcile decision) —
+while every GPU runs the CDNA4 MFMA bf16 GEMM load generator at a high duty
+cycle. Also reports the GPU-util metric error vs rocm-smi (the second
+north-star number) in `config.util_err_pct`.
+
+Reference baseline: the cadence/latency parameters in BASELINE.md — 10 s
+exporter tick, <=30 s to metric availability (no published latency number,
+so vs_baseline is null).
+
+Usage (driver contract):
+    python bench.py [--gpus N] [--steps K] [--warmup W]
+For N>1 the driver launches one rank per GPU via torch.distributed.run;
+ranks read RANK/LOCAL_RANK/WORLD_SIZE from the env. Rank 0 runs the
+exporter + control loop; every rank loads its own GPU (weak scaling: fixed
+per-GPU load). One JSON line on stdout from rank 0.
+"""
+
+import argparse
+import ctypes
+import json
+import os
+import statistics
+import subprocess
+import sys
+import threading
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def rocm_smi_busy():
+    """Per-GPU busy%% straight from rocm-smi (the validation oracle)."""
+    try:
+        out = subprocess.run(
+            ["rocm-smi", "--showuse", "--json"], capture_output=True, timeout=10
+        )
+        data = json.loads(out.stdout.decode())
+        busy = {}
+        for card, vals in data.items():
+            if not card.startswith("card"):
+                continue
+            for k, v in vals.items():
+                if "GPU use" in k:
+                    busy[int(card[4:])] = float(v)
+        return busy
+    except Exception as e:  # noqa: BLE001
+        log(f"rocm-smi unavailable: {e}")
+        return {}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--exporter-interval-ms", type=float, default=100.0)
+    ap.add_argument("--load-util", type=float, default=80.0)
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    has_gpu = torch.cuda.is_available()
+    backend = "nccl" if has_gpu else "gloo"
+    if world > 1:
+        dist.init_process_group(backend=backend)
+    if has_gpu:
+        torch.cuda.set_device(local_rank)
+
+    from mi355x_gpu_hpa import loadgen
+    from mi355x_gpu_hpa.control import (
+        ControlLoop,
+        HpaSpec,
+        Scraper,
+        ScrapeTarget,
+        synth_pod_labels,
+    )
+    from mi355x_gpu_hpa.exporter import ExporterProcess
+
+    # --- start per-rank GPU load: duty-cycled MFMA bf16 GEMM -------------
+    stop_flag = ctypes.c_int(0)
+    load_thread = None
+    if has_gpu:
+        assert loadgen.available(), "HIP loadgen library must be built on a GPU box"
+
+        def burn():
+            lib = loadgen._load()
+            lib.lg_gemm_burn(
+                local_rank, ctypes.c_double(args.load_util),
+                ctypes.c_double(3600.0), 4096, 4096, 4096,
+                ctypes.c_double(50.0), ctypes.byref(stop_flag),
+            )
+
+        load_thread = threading.Thread(target=burn, daemon=True)
+        load_thread.start()
+
+    # --- rank 0: exporter + control loop ---------------------------------
+    exporter = None
+    loop = None
+    util_err_pct = None
+    n_exported = 0
+    if rank == 0:
+        if has_gpu:
+            exporter = ExporterProcess(interval_ms=args.exporter_interval_ms)
+        else:
+            exporter = ExporterProcess(
+                mock_devices=n_gpus, interval_ms=args.exporter_interval_ms
+            )
+        exporter.__enter__()
+        pods = [f"cuda-test-{i}" for i in range(n_gpus)]
+        scraper = Scraper([ScrapeTarget(exporter.url, node="node0")])
+
+        # attach pod identity per GPU the way kubelet attribution would:
+        # post-process scraped samples (gpu index i -> pod cuda-test-i)
+        orig_scrape = scraper.scrape_once
+
+        def scrape_with_pods():
+            samples = orig_scrape()
+            for s in samples:
+                g = s.labels.get("gpu")
+                if g is not None and "pod" not in s.labels:
+                    s.labels["pod"] = f"cuda-test-{g}"
+                    s.labels.setdefault("namespace", "default")
+            return samples
+
+        scraper.scrape_once = scrape_with_pods
+        loop = ControlLoop(
+            scraper,
+            hpa_spec=HpaSpec(min_replicas=1, max_replicas=8, target_value=5.0),
+            extra_samples=lambda: synth_pod_labels(pods),
+        )
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+
+    def sync():
+        if has_gpu:
+            torch.cuda.synchronize()
+
+    # --- warmup -----------------------------------------------------------
+    # let the exporter take >=2 samples so windowed rates exist
+    time.sleep(max(0.5, 2.5 * args.exporter_interval_ms / 1e3))
+    for _ in range(args.warmup):
+        if rank == 0:
+            loop.step()
+    barrier()
+    sync()
+
+    # --- timed region: exactly K control-loop steps -----------------------
+    latencies = []
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        if rank == 0:
+            r = loop.step()
+            latencies.append(r.total_s)
+    barrier()
+    sync()
+    t1 = time.monotonic()
+    wall_s = t1 - t0
+
+    # max over ranks (non-zero ranks have ~0 step time; the max is rank 0's)
+    if world > 1:
+        t = torch.tensor([wall_s])
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        wall_s = float(t[0])
+
+    # --- metric-error validation vs rocm-smi (rank 0, outside timing) ----
+    if rank == 0 and has_gpu:
+        oracle = rocm_smi_busy()
+        samples = __import__(
+            "mi355x_gpu_hpa.control", fromlist=["parse_prometheus_text"]
+        ).parse_prometheus_text(exporter.scrape())
+        ours = {
+            int(s.labels["gpu"]): s.value
+            for s in samples
+            if s.name == "dcgm_gpu_utilization"
+        }
+        n_exported = len(ours)
+        errs = [abs(ours[g] - oracle[g]) for g in ours if g in oracle]
+        if errs:
+            util_err_pct = max(errs)
+            log(f"util err vs rocm-smi: ours={ours} oracle={oracle} "
+                f"max_abs_err={util_err_pct:.1f}%")
+
+    # --- stop load + report ----------------------------------------------
+    stop_flag.value = 1
+    if load_thread:
+        load_thread.join(timeout=10)
+    if exporter:
+        final_replicas = loop.hpa_state.current_replicas if loop else None
+        exporter.__exit__(None, None, None)
+
+    if rank == 0:
+        latencies.sort()
+        p50_ms = statistics.median(latencies) * 1e3
+        p99_ms = latencies[min(len(latencies) - 1, int(len(latencies) * 0.99))] * 1e3
+        result = {
+            "metric": "p50_scrape_to_hpa_scale_latency",
+            "value": p50_ms,
+            "unit": "ms",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": wall_s / args.steps * 1e3,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "mi355x-exporter + mfma-bf16-gemm load",
+                "load": "gemm_bf16 4096^3 duty-cycled",
+                "load_util_target_pct": args.load_util,
+                "exporter_interval_ms": args.exporter_interval_ms,
+                "hpa": "target 5, min 1, max 8 (reference rule semantics)",
+                "parallelism": f"replicas{n_gpus}",
+                "p99_ms": round(p99_ms, 3),
+                "util_err_vs_rocm_smi_pct": util_err_pct,
+                "gpus_exported": n_exported,
+                "final_replicas": final_replicas,
+                "reference_cadence_s": 10.0,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
