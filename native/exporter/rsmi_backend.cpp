@@ -207,6 +207,11 @@ class RsmiBackend final : public Backend {
                 s.mclk_mhz = gm.current_uclk;
             if (u16ok(gm.temperature_mem) && s.temp_mem_c < 0)
                 s.temp_mem_c = gm.temperature_mem;
+            if (u16ok(gm.temperature_edge) && gm.temperature_edge != 0 &&
+                s.temp_edge_c < 0)
+                s.temp_edge_c = gm.temperature_edge;
+            if (u16ok(gm.temperature_hotspot) && s.temp_hotspot_c < 0)
+                s.temp_hotspot_c = gm.temperature_hotspot;
 
             // PCIe: accumulated GB/s-seconds; expose instantaneous if present
             if (gm.pcie_bandwidth_inst && gm.pcie_bandwidth_inst != ~0ull) {
@@ -228,6 +233,10 @@ class RsmiBackend final : public Backend {
             }
             s.xgmi_num_links = nl;
         }
+        // dcgm_gpu_temp must exist (README.md:46 probe): some MI3xx SKUs
+        // expose no edge sensor — fall back to junction/hotspot.
+        if (s.temp_edge_c < 0 && s.temp_hotspot_c >= 0)
+            s.temp_edge_c = s.temp_hotspot_c;
         return s;
     }
 
