@@ -29,6 +29,9 @@ extern "C" __global__ void vector_add_f32(const float*, const float*, float*, in
 extern "C" __global__ void vector_add_f32x4(const float4*, const float4*, float4*, int);
 extern "C" __global__ void gemm_bf16_tn(const unsigned short*, const unsigned short*,
                                         float*, int, int, int, int);
+extern "C" __global__ void gemm_bf16_tn_linear(const unsigned short*,
+                                               const unsigned short*, float*, int,
+                                               int, int, int);
 
 #define LG_CHECK(expr)                                                        \
     do {                                                                      \
@@ -177,20 +180,26 @@ static void gemm_free(GemmBufs& g)
     g = GemmBufs{};
 }
 
-static int gemm_launch(const GemmBufs& g, hipStream_t stream)
+static int gemm_launch(const GemmBufs& g, hipStream_t stream, int variant = 1)
 {
     int n_tiles = (g.m / 128) * (g.n / 128);
     int blocks = n_tiles < 2048 ? n_tiles : 2048;
     int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
-    hipLaunchKernelGGL(gemm_bf16_tn, dim3(blocks), dim3(256), 0, stream,
-                       g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
+    if (variant == 0)
+        hipLaunchKernelGGL(gemm_bf16_tn_linear, dim3(blocks), dim3(256), 0, stream,
+                           g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
+    else
+        hipLaunchKernelGGL(gemm_bf16_tn, dim3(blocks), dim3(256), 0, stream,
+                           g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
     return 0;
 }
 
 // Timed GEMM: `iters` back-to-back launches after `warmup` untimed ones.
 // *ms_out = mean ms per GEMM, *tflops_out = 2*M*N*K / time.
-int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
-                       double* ms_out, double* tflops_out)
+// variant: 1 = st_16x32-swizzled LDS (default), 0 = linear LDS (A/B ref).
+int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
+                               int iters, int variant, double* ms_out,
+                               double* tflops_out)
 {
     if (m % 128 || n % 128 || k % 64) {
         std::snprintf(g_last_error, sizeof(g_last_error),
@@ -200,10 +209,10 @@ int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
-    for (int i = 0; i < warmup; ++i) gemm_launch(g, 0);
+    for (int i = 0; i < warmup; ++i) gemm_launch(g, 0, variant);
     LG_CHECK(hipDeviceSynchronize());
     double t0 = now_ms();
-    for (int i = 0; i < iters; ++i) gemm_launch(g, 0);
+    for (int i = 0; i < iters; ++i) gemm_launch(g, 0, variant);
     LG_CHECK(hipDeviceSynchronize());
     double t1 = now_ms();
     double ms = (t1 - t0) / iters;
@@ -213,10 +222,18 @@ int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
     return 0;
 }
 
+int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
+                       double* ms_out, double* tflops_out)
+{
+    return lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters, 1, ms_out,
+                                      tflops_out);
+}
+
 // Numerics entry: C f32 = A bf16 @ B^T bf16 on caller data (A row-major
 // [m][k] as f32 -> converted; Bt row-major [n][k]).
-int lg_gemm_bf16_verify(int device, const float* a_h, const float* bt_h,
-                        float* c_out, int m, int n, int k)
+// variant: 1 = swizzled (product kernel), 0 = linear LDS.
+int lg_gemm_bf16_verify_variant(int device, const float* a_h, const float* bt_h,
+                                float* c_out, int m, int n, int k, int variant)
 {
     if (m % 128 || n % 128 || k % 64) {
         std::snprintf(g_last_error, sizeof(g_last_error),
@@ -232,12 +249,18 @@ int lg_gemm_bf16_verify(int device, const float* a_h, const float* bt_h,
     tmp.resize((size_t)n * k);
     for (size_t i = 0; i < tmp.size(); ++i) tmp[i] = f32_to_bf16(bt_h[i]);
     LG_CHECK(hipMemcpy(g.bt, tmp.data(), tmp.size() * 2, hipMemcpyHostToDevice));
-    gemm_launch(g, 0);
+    gemm_launch(g, 0, variant);
     LG_CHECK(hipDeviceSynchronize());
     LG_CHECK(hipGetLastError());
     LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
     gemm_free(g);
     return 0;
+}
+
+int lg_gemm_bf16_verify(int device, const float* a_h, const float* bt_h,
+                        float* c_out, int m, int n, int k)
+{
+    return lg_gemm_bf16_verify_variant(device, a_h, bt_h, c_out, m, n, k, 1);
 }
 
 // Duty-cycled GEMM burn: aim at `target_util_pct` GPU-busy for `seconds`.

@@ -24,6 +24,9 @@ int lg_device_count();
 int lg_vector_add_loop(int device, int n, int iters, double* ms_out);
 int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
                        double* ms_out, double* tflops_out);
+int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
+                               int iters, int variant, double* ms_out,
+                               double* tflops_out);
 int lg_gemm_burn(int device, double target_util_pct, double seconds,
                  int m, int n, int k, double period_ms, volatile int* stop_flag);
 }
@@ -67,12 +70,17 @@ int main(int argc, char** argv)
         int k = (int)argd(argc, argv, "--k", 4096);
         int iters = (int)argd(argc, argv, "--iters", 50);
         int warmup = (int)argd(argc, argv, "--warmup", 5);
+        int variant = 1;
+        for (int i = 1; i < argc; ++i)
+            if (!std::strcmp(argv[i], "--linear")) variant = 0;
         double ms = 0, tf = 0;
-        if (lg_gemm_bf16_bench(device, m, n, k, warmup, iters, &ms, &tf)) {
+        if (lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters, variant,
+                                       &ms, &tf)) {
             std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
-        std::printf("gemm_bf16 %dx%dx%d ms=%.3f tflops=%.1f\n", m, n, k, ms, tf);
+        std::printf("gemm_bf16%s %dx%dx%d ms=%.3f tflops=%.1f\n",
+                    variant ? "" : "_linear", m, n, k, ms, tf);
         return 0;
     }
     if (mode == "burn") {
