@@ -111,6 +111,29 @@ def main():
         load_thread = threading.Thread(target=burn, daemon=True)
         load_thread.start()
 
+    # --- xGMI link traffic (world > 1): bucketed RCCL all-reduce ----------
+    # an all-reduce over xGMI is the saturating load that
+    # makes the exporter's amd_xgmi_link_* rates non-zero on the 8-GPU node.
+    # Ring all-reduce is per-link-bandwidth bound on the point-to-point xGMI
+    # topology (7 links x ~153 GB/s) — ideal as a load, not a perf path.
+    # One async all-reduce per step on a dedicated group: collective counts
+    # match on every rank by construction.
+    xgmi_group = None
+    xgmi_buf = None
+    if world > 1:
+        xgmi_group = dist.new_group(backend=backend)
+        xgmi_buf = (
+            torch.empty(32 * 1024 * 1024, dtype=torch.bfloat16, device="cuda")
+            if has_gpu
+            else torch.empty(1024 * 1024, dtype=torch.float32)
+        )
+        xgmi_buf.uniform_()
+
+    def xgmi_tick():
+        if xgmi_group is None:
+            return None
+        return dist.all_reduce(xgmi_buf, group=xgmi_group, async_op=True)
+
     # --- rank 0: exporter + control loop ---------------------------------
     exporter = None
     loop = None
@@ -159,18 +182,25 @@ def main():
     # let the exporter take >=2 samples so windowed rates exist
     time.sleep(max(0.5, 2.5 * args.exporter_interval_ms / 1e3))
     for _ in range(args.warmup):
+        w = xgmi_tick()
         if rank == 0:
             loop.step()
+        if w:
+            w.wait()
     barrier()
     sync()
 
-    # --- timed region: exactly K control-loop steps -----------------------
+    # --- timed region: exactly K control-loop steps, with one xGMI
+    # all-reduce per step when world > 1 -----------------------------------
     latencies = []
     t0 = time.monotonic()
     for _ in range(args.steps):
+        w = xgmi_tick()
         if rank == 0:
             r = loop.step()
             latencies.append(r.total_s)
+        if w:
+            w.wait()
     barrier()
     sync()
     t1 = time.monotonic()
@@ -199,6 +229,10 @@ def main():
             util_err_pct = max(errs)
             log(f"util err vs rocm-smi: ours={ours} oracle={oracle} "
                 f"max_abs_err={util_err_pct:.1f}%")
+        xgmi_bps = [s.value for s in samples
+                    if s.name == "amd_xgmi_total_bytes_per_second"]
+        if xgmi_bps and max(xgmi_bps) > 0:
+            log(f"xGMI total traffic: {max(xgmi_bps)/1e9:.2f} GB/s (max GPU)")
 
     # --- stop load + report ----------------------------------------------
     stop_flag.value = 1
