@@ -65,6 +65,14 @@ def _load():
             np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
         ]
+        lib.lg_gemm_bf16_verify_variant.argtypes = (
+            lib.lg_gemm_bf16_verify.argtypes + [ctypes.c_int]
+        )
+        lib.lg_gemm_bf16_bench_variant.argtypes = [
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double),
+        ]
         lib.lg_gemm_burn.argtypes = [
             ctypes.c_int, ctypes.c_double, ctypes.c_double,
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
@@ -103,24 +111,31 @@ def vector_add(a: np.ndarray, b: np.ndarray, device: int = 0) -> np.ndarray:
     return out
 
 
-def gemm_bf16(a: np.ndarray, bt: np.ndarray, device: int = 0) -> np.ndarray:
-    """C f32 [m,n] = bf16(a) [m,k] @ bf16(bt).T [k,n] on the GPU."""
+def gemm_bf16(a: np.ndarray, bt: np.ndarray, device: int = 0,
+              variant: int = 1) -> np.ndarray:
+    """C f32 [m,n] = bf16(a) [m,k] @ bf16(bt).T [k,n] on the GPU.
+
+    variant: 1 = 128^2 swizzled tile (default), 0 = 128^2 linear LDS,
+    2 = 256^2 8-phase schedule (m,n must be multiples of 256)."""
     a = np.ascontiguousarray(a, np.float32)
     bt = np.ascontiguousarray(bt, np.float32)
     m, k = a.shape
     n, k2 = bt.shape
     assert k == k2
     out = np.empty((m, n), np.float32)
-    _check(_load().lg_gemm_bf16_verify(device, a, bt, out, m, n, k))
+    _check(_load().lg_gemm_bf16_verify_variant(device, a, bt, out, m, n, k,
+                                               variant))
     return out
 
 
-def gemm_bench(m=4096, n=4096, k=4096, warmup=5, iters=50, device=0):
+def gemm_bench(m=4096, n=4096, k=4096, warmup=5, iters=50, device=0,
+               variant=1):
     """Returns (ms_per_gemm, tflops)."""
     ms = ctypes.c_double()
     tf = ctypes.c_double()
-    _check(_load().lg_gemm_bf16_bench(device, m, n, k, warmup, iters,
-                                      ctypes.byref(ms), ctypes.byref(tf)))
+    _check(_load().lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters,
+                                              variant,
+                                              ctypes.byref(ms), ctypes.byref(tf)))
     return ms.value, tf.value
 
 

@@ -1,0 +1,215 @@
+// gemm_bf16_256.hip — 256x256-tile, 8-phase MFMA bf16 GEMM for MI355X
+// (gfx950, CDNA4). The deep-pipelined big-tile schedule: where the 128^2
+// 2-barrier kernel (gemm_bf16.hip) tops out near the structural ~900 TF
+// ceiling (its barrier drains the LDS-DMA queue once per K-step), this one
+// keeps half-tile DMAs in flight across every barrier with counted vmcnt
+// waits and splits each K-step into 4 MFMA phases so staging, LDS reads and
+// matrix math overlap continuously.
+//
+// Geometry (per 512-thread workgroup = 8 waves in a 2(M) x 4(N) grid):
+//   output tile     256 x 256 (wave: 128 x 64 = 8 x 4 fragments of 16x16)
+//   K-step (BK)     64  (2 x mfma_f32_16x16x32_bf16 per fragment)
+//   LDS             128 KiB = 2 buffers x 4 half-tiles x 16 KiB
+//                   half-tiles: A rows 0-127 (A0) / 128-255 (A1),
+//                               B^T cols 0-127 (B0) / 128-255 (B1),
+//                   each a [128][64] bf16 image, st_16x32-swizzled
+//   staging         __builtin_amdgcn_global_load_lds 16 B/lane; one
+//                   half-tile = 2 glds per wave (16 KiB / (8 waves x 1 KiB))
+//
+// Phase schedule per K-tile kt (4 phases, 16 MFMA each; m-half-major so the
+// A fragments read in q0/q1 die early):
+//   q0: m0-3 x n0-1   reads A(m0-3) 8x ds_read_b128 + B(n0-1) 4x
+//   q1: m4-7 x n0-1   reads A(m4-7) 8x               (B regs held)
+//   q2: m0-3 x n2-3   reads A(m0-3) 8x + B(n2-3) 4x
+//   q3: m4-7 x n2-3   reads A(m4-7) 8x               (B regs held)
+//
+// Staging schedule (one half-tile per phase, destination slot provably dead
+// at issue time; src K-tile clamped at the tail — a clamped restage writes
+// byte-identical data, so the overlap is benign):
+//   q0 stages (kt+1, A0)   into the other buffer (occupant kt-1 A0 dead)
+//   q1 stages (kt+2, B0)   into the CURRENT buffer (kt's B0 dead after q0)
+//   q2 stages (kt+1, A1)
+//   q3 stages (kt+1, B1)
+// Per-wave vmcnt drains (before the barrier that publishes them):
+//   end of q3: vmcnt(2)  -> (kt+1) A0,A1,B0 landed for q0/q1 reads
+//   end of q1: vmcnt(4)  -> (kt+1) B1 landed for q2 reads
+//
+// C[M][N] f32 = A[M][K] bf16 @ B^T[N][K] bf16; M,N % 256 == 0, K % 64 == 0.
+
+#include <hip/hip_runtime.h>
+
+typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define HALF_HW 8192          // halfwords per half-tile image (128x64)
+#define HALF_BYTES 16384
+
+static __device__ __forceinline__ int swz256(int byte_off)
+{
+    return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+    const unsigned short* __restrict__ A,  // [M][K] bf16
+    const unsigned short* __restrict__ Bt, // [N][K] bf16
+    float* __restrict__ C,                 // [M][N] f32
+    int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW]; // [buf][half][128][64]
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2; // 0..1: A-half this wave consumes
+    const int wc = w & 3;  // 0..3: B cols wc*64..+64 (B-half = wc>>1)
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles = (M / 256) * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    // glds source mapping for this lane's 16 B of each 1-KiB piece
+    // (piece p = w*2+it covers half-image rows p*8..p*8+7).
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    // LDS read byte offsets (within a half image) for fragment loads.
+    // A-frag mf (0..7), k-step ks: row_in_half = mf*16 + (lane&15)
+    // B-frag nf (0..3): col_in_half = (wc&1)*64 + nf*16 + (lane&15)
+    auto frag_off = [&](int row_in_half, int ks) {
+        return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
+    };
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        const long row0 = (long)(tile / n_tiles_n) * 256;
+        const long col0 = (long)(tile % n_tiles_n) * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        // stage half `h` (0=A0,1=A1,2=B0,3=B1) of K-tile `kt` into buffer
+        // `buf`; 2 glds per wave.
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1; // tail clamp (benign restage)
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        // ---- prologue: kt0's 4 halves + (kt1,B0) [B0 is staged two tiles
+        // ahead by the q1 slot, so kt1's B0 belongs to "kt=-1 q1"] ----
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        stage(1, 2, 1);
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kt0 landed
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[4][2]; // current m-half-range fragments
+        bf16x8 bfrag[2][2]; // current n-pair fragments (held 2 phases)
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned short* la = &lds[(buf * 4 + wr) * HALF_HW];
+            const unsigned short* lb = &lds[(buf * 4 + 2 + (wc >> 1)) * HALF_HW];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = (q & 1) * 4; // q0,q2: m0-3; q1,q3: m4-7
+                const int npair = (q >> 1);    // q0,q1: n0-1; q2,q3: n2-3
+
+                // fragment ds_reads for this phase
+#pragma unroll
+                for (int m = 0; m < 4; ++m) {
+                    const int row = (mbase + m) * 16 + (lane & 15);
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        afrag[m][ks] =
+                            *(const bf16x8*)((const char*)la + frag_off(row, ks));
+                }
+                if ((q & 1) == 0) { // q0/q2: refresh B pair
+#pragma unroll
+                    for (int n = 0; n < 2; ++n) {
+                        const int col = bcol0 + (npair * 2 + n) * 16 + (lane & 15);
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            bfrag[n][ks] =
+                                *(const bf16x8*)((const char*)lb + frag_off(col, ks));
+                    }
+                }
+
+                // staging for this phase (see schedule above)
+                if (q == 0) stage(kt + 1, 0, buf ^ 1);
+                else if (q == 1) stage(kt + 2, 2, buf);
+                else if (q == 2) stage(kt + 1, 1, buf ^ 1);
+                else stage(kt + 1, 3, buf ^ 1);
+
+                __builtin_amdgcn_s_barrier();
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 4; ++m)
+#pragma unroll
+                    for (int n = 0; n < 2; ++n)
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            acc[mbase + m][npair * 2 + n] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag[m][ks], bfrag[n][ks],
+                                    acc[mbase + m][npair * 2 + n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+
+                // per-wave DMA drains, before the barrier that publishes them
+                if (q == 1)
+                    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+                else if (q == 3)
+                    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+                __builtin_amdgcn_s_barrier();
+            }
+        }
+
+        // epilogue: per-wave 128x64 f32 store
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}

@@ -32,6 +32,9 @@ extern "C" __global__ void gemm_bf16_tn(const unsigned short*, const unsigned sh
 extern "C" __global__ void gemm_bf16_tn_linear(const unsigned short*,
                                                const unsigned short*, float*, int,
                                                int, int, int);
+extern "C" __global__ void gemm_bf16_tn_256(const unsigned short*,
+                                            const unsigned short*, float*, int,
+                                            int, int, int);
 
 #define LG_CHECK(expr)                                                        \
     do {                                                                      \
@@ -180,8 +183,17 @@ static void gemm_free(GemmBufs& g)
     g = GemmBufs{};
 }
 
+// variant: 0 = 128^2 linear LDS, 1 = 128^2 swizzled, 2 = 256^2 8-phase.
 static int gemm_launch(const GemmBufs& g, hipStream_t stream, int variant = 1)
 {
+    if (variant == 2) {
+        int n_tiles = (g.m / 256) * (g.n / 256);
+        int blocks = n_tiles < 2048 ? n_tiles : 2048;
+        int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
+        hipLaunchKernelGGL(gemm_bf16_tn_256, dim3(blocks), dim3(512), 0, stream,
+                           g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
+        return 0;
+    }
     int n_tiles = (g.m / 128) * (g.n / 128);
     int blocks = n_tiles < 2048 ? n_tiles : 2048;
     int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
@@ -194,6 +206,17 @@ static int gemm_launch(const GemmBufs& g, hipStream_t stream, int variant = 1)
     return 0;
 }
 
+static bool gemm_dims_ok(int m, int n, int k, int variant)
+{
+    int mt = variant == 2 ? 256 : 128;
+    if (m % mt || n % mt || k % 64) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "gemm dims must be multiples of %d/%d/64", mt, mt);
+        return false;
+    }
+    return true;
+}
+
 // Timed GEMM: `iters` back-to-back launches after `warmup` untimed ones.
 // *ms_out = mean ms per GEMM, *tflops_out = 2*M*N*K / time.
 // variant: 1 = st_16x32-swizzled LDS (default), 0 = linear LDS (A/B ref).
@@ -201,11 +224,7 @@ int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
                                int iters, int variant, double* ms_out,
                                double* tflops_out)
 {
-    if (m % 128 || n % 128 || k % 64) {
-        std::snprintf(g_last_error, sizeof(g_last_error),
-                      "gemm dims must be multiples of 128/128/64");
-        return -1;
-    }
+    if (!gemm_dims_ok(m, n, k, variant)) return -1;
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
@@ -235,11 +254,7 @@ int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
 int lg_gemm_bf16_verify_variant(int device, const float* a_h, const float* bt_h,
                                 float* c_out, int m, int n, int k, int variant)
 {
-    if (m % 128 || n % 128 || k % 64) {
-        std::snprintf(g_last_error, sizeof(g_last_error),
-                      "gemm dims must be multiples of 128/128/64");
-        return -1;
-    }
+    if (!gemm_dims_ok(m, n, k, variant)) return -1;
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, false)) return -1;

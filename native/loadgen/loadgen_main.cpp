@@ -71,16 +71,19 @@ int main(int argc, char** argv)
         int iters = (int)argd(argc, argv, "--iters", 50);
         int warmup = (int)argd(argc, argv, "--warmup", 5);
         int variant = 1;
-        for (int i = 1; i < argc; ++i)
+        for (int i = 1; i < argc; ++i) {
             if (!std::strcmp(argv[i], "--linear")) variant = 0;
+            if (!std::strcmp(argv[i], "--v256")) variant = 2;
+        }
         double ms = 0, tf = 0;
         if (lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters, variant,
                                        &ms, &tf)) {
             std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
+        const char* vname = variant == 0 ? "_linear" : variant == 2 ? "_256" : "";
         std::printf("gemm_bf16%s %dx%dx%d ms=%.3f tflops=%.1f\n",
-                    variant ? "" : "_linear", m, n, k, ms, tf);
+                    vname, m, n, k, ms, tf);
         return 0;
     }
     if (mode == "burn") {

@@ -67,6 +67,39 @@ def test_gemm_bf16_numerics(gpu, m, n, k):
     np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 256, 64), (256, 256, 128),
+                                   (512, 256, 256), (512, 512, 1024)])
+def test_gemm_bf16_256_numerics(gpu, m, n, k):
+    """The 256^2 8-phase kernel vs torch fp32 reference (multi-K-tile shapes
+    exercise the deep staging pipeline; k=64 the clamped tail path)."""
+    import torch
+
+    lg = _loadgen()
+    rng = np.random.default_rng(7)
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
+    a[:, 0] += np.arange(m) * 0.01
+    bt[:, 0] -= np.arange(n) * 0.01
+    got = lg.gemm_bf16(a, bt, variant=2)
+    ref = (torch.from_numpy(a).bfloat16().float()
+           @ torch.from_numpy(bt).bfloat16().float().T).numpy()
+    np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
+
+
+def test_gemm_bf16_256_matches_128(gpu):
+    """Cross-check: both kernels compute identical bf16 sums (same
+    accumulation order over K) — results should agree to fp32 rounding."""
+    lg = _loadgen()
+    rng = np.random.default_rng(8)
+    m = n = 512
+    k = 512
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
+    c1 = lg.gemm_bf16(a, bt, variant=1)
+    c2 = lg.gemm_bf16(a, bt, variant=2)
+    np.testing.assert_allclose(c1, c2, rtol=1e-6, atol=1e-5)
+
+
 def test_gemm_bf16_large_shape(gpu):
     """Grid-stride path: more tiles than CTAs."""
     import torch
