@@ -24,15 +24,17 @@
 //   q3: m4-7 x n2-3   reads A(m4-7) 8x               (B regs held)
 //
 // Staging schedule (one half-tile per phase, destination slot provably dead
-// at issue time; src K-tile clamped at the tail — a clamped restage writes
-// byte-identical data, so the overlap is benign):
+// at issue time; src K-tile clamped at the tail). Note ALL FOUR halves of a
+// K-tile are read at its q0 (each wave's B-half — B0 for wc<2, B1 for wc>=2
+// — serves its n0-1 columns at q0), so a K-tile must be complete before its
+// q0; the across-boundary in-flight half is the NEXT-next tile's B0, whose
+// current-buffer slot is dead after q2:
 //   q0 stages (kt+1, A0)   into the other buffer (occupant kt-1 A0 dead)
-//   q1 stages (kt+2, B0)   into the CURRENT buffer (kt's B0 dead after q0)
-//   q2 stages (kt+1, A1)
-//   q3 stages (kt+1, B1)
-// Per-wave vmcnt drains (before the barrier that publishes them):
-//   end of q3: vmcnt(2)  -> (kt+1) A0,A1,B0 landed for q0/q1 reads
-//   end of q1: vmcnt(4)  -> (kt+1) B1 landed for q2 reads
+//   q1 stages (kt+1, A1)
+//   q2 stages (kt+1, B1)   (occupant kt-1 B1 dead after kt-1 q2)
+//   q3 stages (kt+2, B0)   into the CURRENT buffer (kt's B0 dead after q2)
+// One per-wave vmcnt drain per K-tile, before the barrier that publishes:
+//   end of q3: vmcnt(2) -> all of kt+1 landed; (kt+2,B0) stays in flight
 //
 // C[M][N] f32 = A[M][K] bf16 @ B^T[N][K] bf16; M,N % 256 == 0, K % 64 == 0.
 
@@ -122,7 +124,7 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
         };
 
         // ---- prologue: kt0's 4 halves + (kt1,B0) [B0 is staged two tiles
-        // ahead by the q1 slot, so kt1's B0 belongs to "kt=-1 q1"] ----
+        // ahead by the q3 slot, so kt1's B0 belongs to "kt=-1 q3"] ----
         stage(0, 0, 0);
         stage(0, 1, 0);
         stage(0, 2, 0);
@@ -167,9 +169,9 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
 
                 // staging for this phase (see schedule above)
                 if (q == 0) stage(kt + 1, 0, buf ^ 1);
-                else if (q == 1) stage(kt + 2, 2, buf);
-                else if (q == 2) stage(kt + 1, 1, buf ^ 1);
-                else stage(kt + 1, 3, buf ^ 1);
+                else if (q == 1) stage(kt + 1, 1, buf ^ 1);
+                else if (q == 2) stage(kt + 1, 3, buf ^ 1);
+                else stage(kt + 2, 2, buf);
 
                 __builtin_amdgcn_s_barrier();
                 asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -187,10 +189,8 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
                                     acc[mbase + m][npair * 2 + n], 0, 0, 0);
                 __builtin_amdgcn_s_setprio(0);
 
-                // per-wave DMA drains, before the barrier that publishes them
-                if (q == 1)
-                    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-                else if (q == 3)
+                // per-wave DMA drain, before the barrier that publishes it
+                if (q == 3)
                     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
                 __builtin_amdgcn_s_barrier();
             }
