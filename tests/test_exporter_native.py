@@ -55,6 +55,26 @@ class TestMetricsContract:
         assert len(links) == 7  # 7 xGMI links per MI355X
         assert all("link" in s.labels for s in links)
 
+    def test_xgmi_rate_derivation(self, tmp_path):
+        """Windowed xGMI rates = accumulator delta / dt. The mock backend
+        advances each link's read/write accumulators at busy*10 KB per
+        elapsed ms, so at busy=50 the derived rate must be ~512 MB/s."""
+        busy = tmp_path / "busy"
+        busy.write_text("50\n")
+        with ExporterProcess(mock_devices=1, interval_ms=100,
+                             mock_busy_file=str(busy)) as exp:
+            time.sleep(0.45)  # several windows
+            samples = parse_prometheus_text(exp.scrape())
+        rates = [s.value for s in samples
+                 if s.name == "amd_xgmi_link_read_bytes_per_second"]
+        assert len(rates) == 7
+        expected = 50 * 10 * 1024 * 1000.0  # 512e6 B/s
+        for r in rates:
+            assert abs(r - expected) / expected < 0.25, (r, expected)
+        total = [s.value for s in samples
+                 if s.name == "amd_xgmi_total_bytes_per_second"][0]
+        assert abs(total - expected * 14) / (expected * 14) < 0.25
+
     def test_metric_set_file_filters(self, tmp_path):
         f = tmp_path / "metrics.csv"
         f.write_text("# only two families\ndcgm_gpu_utilization\ndcgm_gpu_temp\n")
