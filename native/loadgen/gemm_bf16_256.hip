@@ -51,7 +51,10 @@ static __device__ __forceinline__ int swz256(int byte_off)
     return byte_off ^ (((byte_off >> 9) & 1) << 5);
 }
 
-extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+// DEPTH: 2 = cross-boundary B0 prefetch (one half in flight across the
+// K-tile boundary); 1 = stage all of kt+1 during kt, full drain per K-tile.
+template <int DEPTH>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl(
     const unsigned short* __restrict__ A,  // [M][K] bf16
     const unsigned short* __restrict__ Bt, // [N][K] bf16
     float* __restrict__ C,                 // [M][N] f32
@@ -129,8 +132,12 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
         stage(0, 1, 0);
         stage(0, 2, 0);
         stage(0, 3, 0);
-        stage(1, 2, 1);
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kt0 landed
+        if (DEPTH == 2) {
+            stage(1, 2, 1);
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kt0 landed
+        } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_s_barrier();
 
         bf16x8 afrag[4][2]; // current m-half-range fragments
@@ -171,7 +178,8 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
                 if (q == 0) stage(kt + 1, 0, buf ^ 1);
                 else if (q == 1) stage(kt + 1, 1, buf ^ 1);
                 else if (q == 2) stage(kt + 1, 3, buf ^ 1);
-                else stage(kt + 2, 2, buf);
+                else if (DEPTH == 2) stage(kt + 2, 2, buf);
+                else stage(kt + 1, 2, buf ^ 1);
 
                 __builtin_amdgcn_s_barrier();
                 asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -190,8 +198,12 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
                 __builtin_amdgcn_s_setprio(0);
 
                 // per-wave DMA drain, before the barrier that publishes it
-                if (q == 3)
-                    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+                if (q == 3) {
+                    if (DEPTH == 2)
+                        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+                    else
+                        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                }
                 __builtin_amdgcn_s_barrier();
             }
         }
@@ -212,4 +224,19 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
         }
         __syncthreads();
     }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<2>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// depth-1 pipeline (full DMA drain per K-tile) — correctness bisect + A/B.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d1(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
