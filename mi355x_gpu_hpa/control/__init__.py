@@ -1,4 +1,12 @@
-from .hpa import HpaSpec, HpaState, desired_replicas, reconcile
+from .hpa import (
+    HpaSpec,
+    HpaState,
+    MetricTarget,
+    desired_replicas,
+    desired_replicas_multi,
+    reconcile,
+    reconcile_multi,
+)
 from .loop import (
     REFERENCE_RULE_EXPR,
     REFERENCE_RULE_NAME,
@@ -11,7 +19,8 @@ from .promql import PromQLError, Sample, evaluate, evaluate_scalar
 from .scraper import Scraper, ScrapeTarget, parse_prometheus_text
 
 __all__ = [
-    "HpaSpec", "HpaState", "desired_replicas", "reconcile",
+    "HpaSpec", "HpaState", "MetricTarget", "desired_replicas",
+    "desired_replicas_multi", "reconcile", "reconcile_multi",
     "ControlLoop", "LoopResult", "RecordingRule", "synth_pod_labels",
     "REFERENCE_RULE_EXPR", "REFERENCE_RULE_NAME",
     "PromQLError", "Sample", "evaluate", "evaluate_scalar",

@@ -57,6 +57,45 @@ def desired_replicas(
     return max(spec.min_replicas, min(spec.max_replicas, desired))
 
 
+@dataclass
+class MetricTarget:
+    """One Object-metric entry of a multi-metric HPA (deploy/multi-metric/
+    cuda-test-hpa-multi.yaml): metric name + target value."""
+
+    name: str
+    target_value: float
+
+
+def desired_replicas_multi(
+    spec: HpaSpec,
+    metrics: List[MetricTarget],
+    current_replicas: int,
+    values: dict,
+) -> int:
+    """Multi-metric desired count: the HPA controller computes a desired
+    replica count per metric and takes the MAX (upstream semantics). A
+    metric with no value contributes the current count (no-change)."""
+    desired = spec.min_replicas
+    any_value = False
+    for m in metrics:
+        v = values.get(m.name)
+        if v is None:
+            d = current_replicas
+        else:
+            any_value = True
+            one = HpaSpec(
+                min_replicas=spec.min_replicas,
+                max_replicas=spec.max_replicas,
+                target_value=m.target_value,
+                tolerance=spec.tolerance,
+            )
+            d = desired_replicas(one, current_replicas, v)
+        desired = max(desired, d)
+    if not any_value:
+        return current_replicas
+    return desired
+
+
 def reconcile(
     spec: HpaSpec, state: HpaState, metric_value: Optional[float], now_s: float
 ) -> int:
@@ -67,6 +106,19 @@ def reconcile(
     window, so a transient dip can't flap the deployment.
     """
     desired = desired_replicas(spec, state.current_replicas, metric_value)
+    return _stabilize(spec, state, desired, now_s)
+
+
+def reconcile_multi(
+    spec: HpaSpec, state: HpaState, metrics: List[MetricTarget],
+    values: dict, now_s: float
+) -> int:
+    """Multi-metric sync (max-of-desireds), with the same stabilization."""
+    desired = desired_replicas_multi(spec, metrics, state.current_replicas, values)
+    return _stabilize(spec, state, desired, now_s)
+
+
+def _stabilize(spec: HpaSpec, state: HpaState, desired: int, now_s: float) -> int:
     state.recommendations.append((now_s, desired))
     cutoff = now_s - spec.downscale_stabilization_s
     state.recommendations = [(t, d) for (t, d) in state.recommendations if t >= cutoff]

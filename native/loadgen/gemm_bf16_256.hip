@@ -132,7 +132,13 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
         stage(0, 1, 0);
         stage(0, 2, 0);
         stage(0, 3, 0);
-        if (DEPTH == 2) {
+        if (DEPTH == 4) {
+            // B halves are staged two K-tiles ahead (B frags all read at
+            // q0, so B slots die a whole tile early)
+            stage(1, 2, 1);
+            stage(1, 3, 1);
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); // kt0 landed
+        } else if (DEPTH == 2) {
             stage(1, 2, 1);
             asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kt0 landed
         } else {
@@ -141,7 +147,8 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
         __builtin_amdgcn_s_barrier();
 
         bf16x8 afrag[4][2]; // current m-half-range fragments
-        bf16x8 bfrag[2][2]; // current n-pair fragments (held 2 phases)
+        bf16x8 bfrag[4][2]; // n fragments (DEPTH==4: all four held from q0;
+                            // otherwise the current pair, held 2 phases)
 
         for (int kt = 0; kt < kTiles; ++kt) {
             const int buf = kt & 1;
@@ -163,7 +170,18 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                         afrag[m][ks] =
                             *(const bf16x8*)((const char*)la + frag_off(row, ks));
                 }
-                if ((q & 1) == 0) { // q0/q2: refresh B pair
+                if (DEPTH == 4) { // all four B frags read at q0, held to q3
+                    if (q == 0) {
+#pragma unroll
+                        for (int n = 0; n < 4; ++n) {
+                            const int col = bcol0 + n * 16 + (lane & 15);
+#pragma unroll
+                            for (int ks = 0; ks < 2; ++ks)
+                                bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                                frag_off(col, ks));
+                        }
+                    }
+                } else if ((q & 1) == 0) { // q0/q2: refresh B pair
 #pragma unroll
                     for (int n = 0; n < 2; ++n) {
                         const int col = bcol0 + (npair * 2 + n) * 16 + (lane & 15);
@@ -175,7 +193,19 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                 }
 
                 // staging for this phase (see schedule above)
-                if (q == 0) stage(kt + 1, 0, buf ^ 1);
+                if (DEPTH == 4) {
+                    // A halves one tile ahead at q0; B halves TWO tiles
+                    // ahead at q1/q2 (their current-buffer slots are dead
+                    // after q0's reads)
+                    if (q == 0) {
+                        stage(kt + 1, 0, buf ^ 1);
+                        stage(kt + 1, 1, buf ^ 1);
+                    } else if (q == 1) {
+                        stage(kt + 2, 2, buf);
+                    } else if (q == 2) {
+                        stage(kt + 2, 3, buf);
+                    }
+                } else if (q == 0) stage(kt + 1, 0, buf ^ 1);
                 else if (q == 1) stage(kt + 1, 1, buf ^ 1);
                 else if (q == 2) stage(kt + 1, 3, buf ^ 1);
                 else if (DEPTH == 2) stage(kt + 2, 2, buf);
@@ -193,13 +223,16 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                         for (int ks = 0; ks < 2; ++ks)
                             acc[mbase + m][npair * 2 + n] =
                                 __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    afrag[m][ks], bfrag[n][ks],
+                                    afrag[m][ks],
+                                    bfrag[DEPTH == 4 ? npair * 2 + n : n][ks],
                                     acc[mbase + m][npair * 2 + n], 0, 0, 0);
                 __builtin_amdgcn_s_setprio(0);
 
                 // per-wave DMA drain, before the barrier that publishes it
                 if (q == 3) {
-                    if (DEPTH == 2)
+                    if (DEPTH == 4)
+                        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+                    else if (DEPTH == 2)
                         asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
                     else
                         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -239,4 +272,14 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d1(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// deep variant: B halves staged two K-tiles ahead, all B frags held from q0
+// (2 half-tiles in flight across the boundary; >=4 phases of DMA latency
+// budget per half).
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d4(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<4>(A, Bt, C, M, N, K, tiles_per_cta);
 }

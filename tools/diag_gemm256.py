@@ -21,7 +21,7 @@ def check_shapes(m, n, k):
     abf = torch.from_numpy(a).bfloat16().float().numpy()
     btbf = torch.from_numpy(bt).bfloat16().float().numpy()
     ref = abf @ btbf.T
-    for variant in (2, 3):
+    for variant in (2, 3, 4):
         got = loadgen.gemm_bf16(a, bt, variant=variant)
         bad = np.abs(got - ref) > 0.5
         print(f"variant {variant} ({m}x{n}x{k}): mismatch {bad.sum()}/{bad.size}"
