@@ -1,7 +1,15 @@
 """Unit tests for the HPA reconcile algorithm (reference semantics:
 cuda-test-hpa.yaml:11-21, desired = ceil(current/target x replicas))."""
 
-from mi355x_gpu_hpa.control import HpaSpec, HpaState, desired_replicas, reconcile
+from mi355x_gpu_hpa.control import (
+    HpaSpec,
+    HpaState,
+    MetricTarget,
+    desired_replicas,
+    desired_replicas_multi,
+    reconcile,
+    reconcile_multi,
+)
 
 
 def spec(**kw):
@@ -59,6 +67,33 @@ class TestReconcile:
         assert st.current_replicas == 3
         assert reconcile(s, st, 4.9, now_s=10.0) == 3   # dip within window
         assert reconcile(s, st, 10.0, now_s=20.0) == 3  # back up
+
+    def test_multi_metric_max_wins(self):
+        # deploy/multi-metric/cuda-test-hpa-multi.yaml semantics: desired =
+        # max over metrics; a bandwidth-bound load scales on HBM BW before
+        # busy% trips
+        s = spec(max_replicas=8)
+        metrics = [MetricTarget("cuda_test_gpu_avg", 40.0),
+                   MetricTarget("cuda_test_hbm_bw_avg", 60.0)]
+        # busy below target, bandwidth way above
+        d = desired_replicas_multi(s, metrics, 1,
+                                   {"cuda_test_gpu_avg": 30.0,
+                                    "cuda_test_hbm_bw_avg": 95.0})
+        assert d == 2  # ceil(95/60) = 2 > ceil(30/40)=1(tolerance band aside)
+
+    def test_multi_metric_missing_one(self):
+        s = spec(max_replicas=8)
+        metrics = [MetricTarget("a", 10.0), MetricTarget("b", 10.0)]
+        assert desired_replicas_multi(s, metrics, 2, {"a": None, "b": None}) == 2
+        assert desired_replicas_multi(s, metrics, 2, {"a": 30.0}) == 6
+
+    def test_reconcile_multi_stabilized(self):
+        s = spec(max_replicas=8, downscale_stabilization_s=60)
+        st = HpaState(current_replicas=1)
+        metrics = [MetricTarget("a", 10.0)]
+        assert reconcile_multi(s, st, metrics, {"a": 50.0}, now_s=0.0) == 5
+        assert reconcile_multi(s, st, metrics, {"a": 0.0}, now_s=10.0) == 5
+        assert reconcile_multi(s, st, metrics, {"a": 0.0}, now_s=61.0) == 1
 
     def test_scale_up_curve_to_eight(self):
         # config 4: 1 -> 8 replica scale-up under sustained high load
