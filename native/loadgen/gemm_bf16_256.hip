@@ -138,7 +138,7 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
             stage(1, 2, 1);
             stage(1, 3, 1);
             asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); // kt0 landed
-        } else if (DEPTH == 2) {
+        } else if (DEPTH == 2 || DEPTH == 5) {
             stage(1, 2, 1);
             asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // kt0 landed
         } else {
@@ -205,6 +205,17 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                     } else if (q == 2) {
                         stage(kt + 2, 3, buf);
                     }
+                } else if (DEPTH == 5) {
+                    // latency-balanced: both A halves at q0, B1 at q1 — the
+                    // last-staged kt+1 half gets 3 phases to land
+                    if (q == 0) {
+                        stage(kt + 1, 0, buf ^ 1);
+                        stage(kt + 1, 1, buf ^ 1);
+                    } else if (q == 1) {
+                        stage(kt + 1, 3, buf ^ 1);
+                    } else if (q == 3) {
+                        stage(kt + 2, 2, buf);
+                    }
                 } else if (q == 0) stage(kt + 1, 0, buf ^ 1);
                 else if (q == 1) stage(kt + 1, 1, buf ^ 1);
                 else if (q == 2) stage(kt + 1, 3, buf ^ 1);
@@ -232,7 +243,7 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                 if (q == 3) {
                     if (DEPTH == 4)
                         asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-                    else if (DEPTH == 2)
+                    else if (DEPTH == 2 || DEPTH == 5)
                         asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
                     else
                         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -282,4 +293,12 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d4(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl<4>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// latency-balanced staging order (A0+A1 at q0, B1 at q1, B0 two ahead).
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d5(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<5>(A, Bt, C, M, N, K, tiles_per_cta);
 }
