@@ -132,7 +132,7 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
         stage(0, 1, 0);
         stage(0, 2, 0);
         stage(0, 3, 0);
-        if (DEPTH == 4) {
+        if (DEPTH == 4 || DEPTH == 6) {
             // B halves are staged two K-tiles ahead (B frags all read at
             // q0, so B slots die a whole tile early)
             stage(1, 2, 1);
@@ -158,19 +158,24 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
 
 #pragma unroll
             for (int q = 0; q < 4; ++q) {
-                const int mbase = (q & 1) * 4; // q0,q2: m0-3; q1,q3: m4-7
+                // DEPTH 6: phase q = m-frags {2q, 2q+1} x all n (2 A reads
+                // held per phase, all 4 B frags held from q0 — 16 LDS reads
+                // per K-tile instead of 40). Others: m-half-major quadrants.
+                const int mbase = DEPTH == 6 ? q * 2 : (q & 1) * 4;
                 const int npair = (q >> 1);    // q0,q1: n0-1; q2,q3: n2-3
+                const int n_m = DEPTH == 6 ? 2 : 4;
 
                 // fragment ds_reads for this phase
 #pragma unroll
                 for (int m = 0; m < 4; ++m) {
+                    if (m >= n_m) break;
                     const int row = (mbase + m) * 16 + (lane & 15);
 #pragma unroll
                     for (int ks = 0; ks < 2; ++ks)
                         afrag[m][ks] =
                             *(const bf16x8*)((const char*)la + frag_off(row, ks));
                 }
-                if (DEPTH == 4) { // all four B frags read at q0, held to q3
+                if (DEPTH == 4 || DEPTH == 6) { // all four B frags at q0
                     if (q == 0) {
 #pragma unroll
                         for (int n = 0; n < 4; ++n) {
@@ -193,7 +198,19 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                 }
 
                 // staging for this phase (see schedule above)
-                if (DEPTH == 4) {
+                if (DEPTH == 6) {
+                    // A halves one tile ahead at q0 (slots dead since kt-1
+                    // end); B halves TWO tiles ahead at q1/q2 (B slots dead
+                    // after q0's reads)
+                    if (q == 0) {
+                        stage(kt + 1, 0, buf ^ 1);
+                        stage(kt + 1, 1, buf ^ 1);
+                    } else if (q == 1) {
+                        stage(kt + 2, 2, buf);
+                    } else if (q == 2) {
+                        stage(kt + 2, 3, buf);
+                    }
+                } else if (DEPTH == 4) {
                     // A halves one tile ahead at q0; B halves TWO tiles
                     // ahead at q1/q2 (their current-buffer slots are dead
                     // after q0's reads)
@@ -226,22 +243,35 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                 asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
                 __builtin_amdgcn_s_setprio(1);
+                if (DEPTH == 6) {
 #pragma unroll
-                for (int m = 0; m < 4; ++m)
+                    for (int m = 0; m < 2; ++m)
 #pragma unroll
-                    for (int n = 0; n < 2; ++n)
+                        for (int n = 0; n < 4; ++n)
 #pragma unroll
-                        for (int ks = 0; ks < 2; ++ks)
-                            acc[mbase + m][npair * 2 + n] =
-                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    afrag[m][ks],
-                                    bfrag[DEPTH == 4 ? npair * 2 + n : n][ks],
-                                    acc[mbase + m][npair * 2 + n], 0, 0, 0);
+                            for (int ks = 0; ks < 2; ++ks)
+                                acc[mbase + m][n] =
+                                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                        afrag[m][ks], bfrag[n][ks],
+                                        acc[mbase + m][n], 0, 0, 0);
+                } else {
+#pragma unroll
+                    for (int m = 0; m < 4; ++m)
+#pragma unroll
+                        for (int n = 0; n < 2; ++n)
+#pragma unroll
+                            for (int ks = 0; ks < 2; ++ks)
+                                acc[mbase + m][npair * 2 + n] =
+                                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                        afrag[m][ks],
+                                        bfrag[DEPTH == 4 ? npair * 2 + n : n][ks],
+                                        acc[mbase + m][npair * 2 + n], 0, 0, 0);
+                }
                 __builtin_amdgcn_s_setprio(0);
 
                 // per-wave DMA drain, before the barrier that publishes it
                 if (q == 3) {
-                    if (DEPTH == 4)
+                    if (DEPTH == 4 || DEPTH == 6)
                         asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
                     else if (DEPTH == 2 || DEPTH == 5)
                         asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
@@ -301,4 +331,13 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d5(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl<5>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// m-quarter phases + all-B-held: 16 LDS reads per K-tile, B staged two
+// K-tiles ahead, same frag-register footprint as the d2 schedule.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d6(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<6>(A, Bt, C, M, N, K, tiles_per_cta);
 }
