@@ -16,25 +16,24 @@
 //   staging         __builtin_amdgcn_global_load_lds 16 B/lane; one
 //                   half-tile = 2 glds per wave (16 KiB / (8 waves x 1 KiB))
 //
-// Phase schedule per K-tile kt (4 phases, 16 MFMA each; m-half-major so the
-// A fragments read in q0/q1 die early):
-//   q0: m0-3 x n0-1   reads A(m0-3) 8x ds_read_b128 + B(n0-1) 4x
-//   q1: m4-7 x n0-1   reads A(m4-7) 8x               (B regs held)
-//   q2: m0-3 x n2-3   reads A(m0-3) 8x + B(n2-3) 4x
-//   q3: m4-7 x n2-3   reads A(m4-7) 8x               (B regs held)
+// PRODUCT phase schedule (DEPTH 6, measured best — profiles/): phase q
+// computes m-frags {2q, 2q+1} x ALL FOUR n-frags (16 MFMA):
+//   q0 reads: A frags 0-1 (4x ds_read_b128) + all B frags (8x), B held to q3
+//   q1-q3 reads: 4x A each  -> 16 LDS reads per K-tile (vs 40 for the
+//   quadrant schedules), and every B slot is dead after q0, which lets B
+//   halves stage TWO K-tiles ahead:
+//   q0 stages (kt+1, A0) + (kt+1, A1) into the other buffer
+//   q1 stages (kt+2, B0) into the CURRENT buffer (B0 dead after q0)
+//   q2 stages (kt+2, B1) into the CURRENT buffer
+//   one per-wave drain at q3: vmcnt(4) -> all of kt+1 landed, kt+2's B
+//   halves stay in flight across the boundary. Every half-tile gets >=4
+//   phases of DMA-latency budget.
 //
-// Staging schedule (one half-tile per phase, destination slot provably dead
-// at issue time; src K-tile clamped at the tail). Note ALL FOUR halves of a
-// K-tile are read at its q0 (each wave's B-half — B0 for wc<2, B1 for wc>=2
-// — serves its n0-1 columns at q0), so a K-tile must be complete before its
-// q0; the across-boundary in-flight half is the NEXT-next tile's B0, whose
-// current-buffer slot is dead after q2:
-//   q0 stages (kt+1, A0)   into the other buffer (occupant kt-1 A0 dead)
-//   q1 stages (kt+1, A1)
-//   q2 stages (kt+1, B1)   (occupant kt-1 B1 dead after kt-1 q2)
-//   q3 stages (kt+2, B0)   into the CURRENT buffer (kt's B0 dead after q2)
-// One per-wave vmcnt drain per K-tile, before the barrier that publishes:
-//   end of q3: vmcnt(2) -> all of kt+1 landed; (kt+2,B0) stays in flight
+// Measured ablations kept as _d* entry points (same template, different
+// DEPTH): d1 full-drain depth-1; d2 half-per-phase with one B0 in flight
+// (846-880 TF); d4 quadrant phases + all-B-held (register spills, 666 TF);
+// d5 latency-balanced d2 (==d2). Src K-tiles are clamped at the tail (a
+// clamped restage writes byte-identical data, so the overlap is benign).
 //
 // C[M][N] f32 = A[M][K] bf16 @ B^T[N][K] bf16; M,N % 256 == 0, K % 64 == 0.
 
@@ -300,7 +299,18 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
     }
 }
 
+// PRODUCT kernel: the d6 schedule (m-quarter phases, all-B-held, B staged
+// two K-tiles ahead) — measured 946-955 TF/s @8192^3, 1033 TF/s @16k x 16k
+// x 8k on random bf16 (see profiles/gemm_bf16_256_ladder.md).
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<6>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// the earlier half-per-phase schedule (one B0 half in flight) — ablation.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d2(
     const unsigned short* A, const unsigned short* Bt, float* C,
     int M, int N, int K, int tiles_per_cta)
 {
@@ -331,13 +341,4 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d5(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl<5>(A, Bt, C, M, N, K, tiles_per_cta);
-}
-
-// m-quarter phases + all-B-held: 16 LDS reads per K-tile, B staged two
-// K-tiles ahead, same frag-register footprint as the d2 schedule.
-extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d6(
-    const unsigned short* A, const unsigned short* Bt, float* C,
-    int M, int N, int K, int tiles_per_cta)
-{
-    gemm_bf16_tn_256_impl<6>(A, Bt, C, M, N, K, tiles_per_cta);
 }
