@@ -22,13 +22,14 @@ HttpServer::~HttpServer() { stop(); }
 
 bool HttpServer::start(std::string* err)
 {
-    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
-    if (listen_fd_ < 0) {
+    int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+    listen_fd_ = fd;
+    if (fd < 0) {
         if (err) *err = "socket() failed";
         return false;
     }
     int one = 1;
-    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
     sockaddr_in addr;
     std::memset(&addr, 0, sizeof(addr));
     addr.sin_family = AF_INET;
@@ -37,22 +38,25 @@ bool HttpServer::start(std::string* err)
         addr.sin_addr.s_addr = INADDR_ANY;
     else if (inet_pton(AF_INET, bind_addr_.c_str(), &addr.sin_addr) != 1) {
         if (err) *err = "bad bind address " + bind_addr_;
-        ::close(listen_fd_);
+        ::close(fd);
+        listen_fd_ = -1;
         return false;
     }
-    if (::bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0) {
+    if (::bind(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
         if (err) *err = "bind failed on port " + std::to_string(port_);
-        ::close(listen_fd_);
+        ::close(fd);
+        listen_fd_ = -1;
         return false;
     }
     if (port_ == 0) {
         socklen_t len = sizeof(addr);
-        getsockname(listen_fd_, (sockaddr*)&addr, &len);
+        getsockname(fd, (sockaddr*)&addr, &len);
         port_ = ntohs(addr.sin_port);
     }
-    if (::listen(listen_fd_, 64) != 0) {
+    if (::listen(fd, 64) != 0) {
         if (err) *err = "listen failed";
-        ::close(listen_fd_);
+        ::close(fd);
+        listen_fd_ = -1;
         return false;
     }
     stop_ = false;
@@ -63,10 +67,14 @@ bool HttpServer::start(std::string* err)
 void HttpServer::stop()
 {
     stop_ = true;
-    if (listen_fd_ >= 0) {
-        ::shutdown(listen_fd_, SHUT_RDWR);
-        ::close(listen_fd_);
-        listen_fd_ = -1;
+    int fd = listen_fd_.exchange(-1);
+    if (fd >= 0) {
+        ::shutdown(fd, SHUT_RDWR);
+        // close AFTER joining the accept loop so its poll/accept never
+        // touches a recycled fd number
+        if (thread_.joinable()) thread_.join();
+        ::close(fd);
+        return;
     }
     if (thread_.joinable()) thread_.join();
 }
@@ -74,11 +82,13 @@ void HttpServer::stop()
 void HttpServer::accept_loop()
 {
     while (!stop_) {
-        pollfd p{listen_fd_, POLLIN, 0};
+        int lfd = listen_fd_.load();
+        if (lfd < 0) break;
+        pollfd p{lfd, POLLIN, 0};
         int rc = ::poll(&p, 1, 250);
         if (stop_) break;
         if (rc <= 0) continue;
-        int fd = ::accept(listen_fd_, nullptr, nullptr);
+        int fd = ::accept(lfd, nullptr, nullptr);
         if (fd < 0) continue;
         // handle inline: requests are tiny and render is fast; a stuck
         // client can't stall us thanks to the send timeout.

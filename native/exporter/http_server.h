@@ -41,7 +41,7 @@ class HttpServer {
     int port_;
     Handler metrics_;
     ReadyFn ready_;
-    int listen_fd_ = -1;
+    std::atomic<int> listen_fd_{-1};
     std::atomic<bool> stop_{false};
     std::thread thread_;
 };
