@@ -342,3 +342,176 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d5(
 {
     gemm_bf16_tn_256_impl<5>(A, Bt, C, M, N, K, tiles_per_cta);
 }
+
+// ---------------------------------------------------------------------------
+// d7: the d6 schedule with 5 barriers per K-tile instead of 8.
+//
+// PMC on d6 shows 34% of wave time parked at barriers/waitcnts. Mid-tile,
+// every fragment read targets data published at the TILE boundary (the
+// per-phase publishes only matter for the next tile), so the closing
+// barrier of each phase is unnecessary if each phase drains its LDS reads
+// (lgkmcnt(0)) BEFORE its single barrier — barrier passage then proves all
+// waves' reads retired, which is exactly the deadness proof the next
+// phase's glds staging needs. Phase body becomes
+//     reads -> lgkmcnt(0) -> s_barrier -> MFMA -> stage
+// with one extra boundary barrier per K-tile after the q3 vmcnt drain
+// (the only point where freshly landed halves must be published to reads).
+// A fast wave's next-phase reads now overlap a slow wave's MFMA segment.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void gemm_bf16_tn_256_impl7(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;
+    const int wc = w & 3;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles = (M / 256) * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    auto frag_off = [&](int row_in_half, int ks) {
+        return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
+    };
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        const long row0 = (long)(tile / n_tiles_n) * 256;
+        const long col0 = (long)(tile % n_tiles_n) * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        stage(1, 2, 1);
+        stage(1, 3, 1);
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[2][2];
+        bf16x8 bfrag[4][2];
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned short* la = &lds[(buf * 4 + wr) * HALF_HW];
+            const unsigned short* lb = &lds[(buf * 4 + 2 + (wc >> 1)) * HALF_HW];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = q * 2;
+
+                // reads, drained BEFORE the phase barrier
+#pragma unroll
+                for (int m = 0; m < 2; ++m) {
+                    const int row = (mbase + m) * 16 + (lane & 15);
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        afrag[m][ks] =
+                            *(const bf16x8*)((const char*)la + frag_off(row, ks));
+                }
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 4; ++n) {
+                        const int col = bcol0 + n * 16 + (lane & 15);
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                            frag_off(col, ks));
+                    }
+                }
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                __builtin_amdgcn_s_barrier();
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            acc[mbase + m][n] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag[m][ks], bfrag[n][ks],
+                                    acc[mbase + m][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 2, 2, buf);
+                } else if (q == 2) {
+                    stage(kt + 2, 3, buf);
+                }
+            }
+            // tile boundary: land kt+1, publish to the next tile's reads
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d7(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl7(A, Bt, C, M, N, K, tiles_per_cta);
+}
