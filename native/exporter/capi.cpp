@@ -11,7 +11,10 @@
 #include "prom_render.h"
 #include "sampler.h"
 
+#include <algorithm>
+#include <chrono>
 #include <cstring>
+#include <vector>
 #include <sstream>
 #include <string>
 
@@ -101,6 +104,38 @@ int mi355x_render_mock_metrics(int n_devices, const char* attr_key,
     }
     return copy_out(mi355x::render_metrics(sampler.snapshot(), attr, opt), buf,
                     buflen);
+}
+
+// Benchmark the real rsmi sampling path: n full-device sampling passes.
+// Returns 0 and fills stats (mean/p50/max microseconds per pass + device
+// count), or -1 with the error in errbuf (no GPU / no rocm_smi).
+int mi355x_sample_benchmark(int n, double* mean_us, double* p50_us,
+                            double* max_us, int* n_devices, char* errbuf,
+                            int errlen)
+{
+    std::string err;
+    auto backend = mi355x::make_rsmi_backend(&err);
+    if (!backend) {
+        std::snprintf(errbuf, errlen, "%s", err.c_str());
+        return -1;
+    }
+    mi355x::Sampler sampler(backend.get(), 1e9);
+    std::vector<double> us;
+    us.reserve(n);
+    for (int i = 0; i < n; ++i) {
+        auto t0 = std::chrono::steady_clock::now();
+        sampler.sample_once();
+        auto t1 = std::chrono::steady_clock::now();
+        us.push_back(std::chrono::duration<double, std::micro>(t1 - t0).count());
+    }
+    std::sort(us.begin(), us.end());
+    double sum = 0;
+    for (double v : us) sum += v;
+    if (mean_us) *mean_us = sum / us.size();
+    if (p50_us) *p50_us = us[us.size() / 2];
+    if (max_us) *max_us = us.back();
+    if (n_devices) *n_devices = backend->device_count();
+    return 0;
 }
 
 } // extern "C"
