@@ -322,7 +322,7 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
     const unsigned short* A, const unsigned short* Bt, float* C,
     int M, int N, int K, int tiles_per_cta)
 {
-    gemm_bf16_tn_256_impl<6>(A, Bt, C, M, N, K, tiles_per_cta);
+    gemm_bf16_tn_256_impl<6, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
 // the earlier half-per-phase schedule (one B0 half in flight) — ablation.
@@ -532,10 +532,11 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d7(
     gemm_bf16_tn_256_impl7(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
-// d8 = product d6 schedule + 4x4 super-tile rasterization.
+// d8 = the d6 schedule WITHOUT super-tile rasterization (raster ablation;
+// the product entry above includes it: +6%% @8192^3, fetch 4.4->? GB).
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d8(
     const unsigned short* A, const unsigned short* Bt, float* C,
     int M, int N, int K, int tiles_per_cta)
 {
-    gemm_bf16_tn_256_impl<6, 1>(A, Bt, C, M, N, K, tiles_per_cta);
+    gemm_bf16_tn_256_impl<6, 0>(A, Bt, C, M, N, K, tiles_per_cta);
 }

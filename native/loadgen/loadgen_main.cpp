@@ -75,6 +75,8 @@ int main(int argc, char** argv)
             if (!std::strcmp(argv[i], "--linear")) variant = 0;
             if (!std::strcmp(argv[i], "--v256")) variant = 2;
         }
+        int vflag = (int)argd(argc, argv, "--variant", -1);
+        if (vflag >= 0) variant = vflag;
         double ms = 0, tf = 0;
         if (lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters, variant,
                                        &ms, &tf)) {
