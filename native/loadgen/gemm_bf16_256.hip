@@ -374,6 +374,7 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d5(
 // (the only point where freshly landed halves must be published to reads).
 // A fast wave's next-phase reads now overlap a slow wave's MFMA segment.
 // ---------------------------------------------------------------------------
+template <int RASTER = 0>
 __device__ __forceinline__ void gemm_bf16_tn_256_impl7(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
@@ -406,11 +407,24 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl7(
         return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
     };
 
+    const int n_tiles_m = M / 256;
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
     for (int t = 0; t < tiles_per_cta; ++t) {
         const int tile = wgid + t * nwg;
         if (tile >= n_tiles) return;
-        const long row0 = (long)(tile / n_tiles_n) * 256;
-        const long col0 = (long)(tile % n_tiles_n) * 256;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
 
         f32x4 acc[8][4];
 #pragma unroll
@@ -529,7 +543,7 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d7(
     const unsigned short* A, const unsigned short* Bt, float* C,
     int M, int N, int K, int tiles_per_cta)
 {
-    gemm_bf16_tn_256_impl7(A, Bt, C, M, N, K, tiles_per_cta);
+    gemm_bf16_tn_256_impl7<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
 // d8 = the d6 schedule WITHOUT super-tile rasterization (raster ablation;
