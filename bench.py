@@ -208,7 +208,7 @@ def main():
 
     # max over ranks (non-zero ranks have ~0 step time; the max is rank 0's)
     if world > 1:
-        t = torch.tensor([wall_s])
+        t = torch.tensor([wall_s], device="cuda" if backend == "nccl" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         wall_s = float(t[0])
 
