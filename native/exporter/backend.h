@@ -60,6 +60,19 @@ struct GpuSample {
     // grained than busy_pct on some firmware)
     double gfx_activity_pct = -1;
     double umc_activity_pct = -1;
+
+    // RAS: accumulated ECC error counts summed over GPU blocks
+    double ecc_correctable = -1;
+    double ecc_uncorrectable = -1;
+
+    // throttle residency accumulators (gpu_metrics v1.6): PVIOL/TVIOL % =
+    // delta(residency)/delta(accumulation_counter) * 100 over the window
+    double accumulation_counter = -1;
+    double ppt_residency_acc = -1;
+    double thm_residency_acc = -1;
+
+    // PCIe replay events (accumulated)
+    double pcie_replay_count = -1;
 };
 
 struct GpuInfo {

@@ -91,6 +91,13 @@ class MockBackend final : public Backend {
         // busy-time accumulators: ~busy% of wall, in ms units
         s.gfx_activity_acc = elapsed * s.busy_pct / 100.0;
         s.mem_activity_acc = elapsed * s.mem_busy_pct / 100.0;
+        // RAS / throttle / replay synthetics
+        s.ecc_correctable = dev;       // stable per-device counter
+        s.ecc_uncorrectable = 0;
+        s.accumulation_counter = elapsed;        // 1 unit per ms
+        s.ppt_residency_acc = elapsed * 0.10;    // 10% PVIOL
+        s.thm_residency_acc = elapsed * 0.02;    // 2% TVIOL
+        s.pcie_replay_count = 3;
         return s;
     }
 

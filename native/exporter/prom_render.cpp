@@ -137,6 +137,29 @@ std::string render_metrics(const std::vector<DeviceMetrics>& devs,
     for (auto& d : devs)
         w.sample(d, attr, d.sample.pcie_rx_bps < 0 ? -1 : d.sample.pcie_rx_bps / 1e3);
 
+    w.family("dcgm_ecc_sbe_aggregate_total",
+             "Accumulated correctable (single-bit) ECC errors, all blocks.",
+             "counter");
+    for (auto& d : devs) w.sample(d, attr, d.sample.ecc_correctable);
+
+    w.family("dcgm_ecc_dbe_aggregate_total",
+             "Accumulated uncorrectable (double-bit) ECC errors, all blocks.",
+             "counter");
+    for (auto& d : devs) w.sample(d, attr, d.sample.ecc_uncorrectable);
+
+    w.family("dcgm_pcie_replay_counter", "Accumulated PCIe replay events.",
+             "counter");
+    for (auto& d : devs) w.sample(d, attr, d.sample.pcie_replay_count);
+
+    w.family("dcgm_power_violation",
+             "Power-throttle (PPT) residency %% over the sampling window.",
+             "gauge");
+    for (auto& d : devs) w.sample(d, attr, d.pviol_pct);
+
+    w.family("dcgm_thermal_violation",
+             "Thermal-throttle residency %% over the sampling window.", "gauge");
+    for (auto& d : devs) w.sample(d, attr, d.tviol_pct);
+
     // --- MI355X-native amd_* families (no reference counterpart) ---
     w.family("amd_gpu_busy_percent_windowed",
              "GPU busy % derived from the gfx activity accumulator over the "

@@ -49,6 +49,7 @@ struct RsmiApi {
     DECL(rsmi_dev_energy_count_get);
     DECL(rsmi_dev_gpu_clk_freq_get);
     DECL(rsmi_dev_gpu_metrics_info_get);
+    DECL(rsmi_dev_ecc_count_get);
 #undef DECL
 
     bool load(std::string* err)
@@ -85,6 +86,7 @@ struct RsmiApi {
         RESOLVE(rsmi_dev_energy_count_get)
         RESOLVE(rsmi_dev_gpu_clk_freq_get)
         RESOLVE(rsmi_dev_gpu_metrics_info_get)
+        RESOLVE(rsmi_dev_ecc_count_get)
 #undef RESOLVE
         return true;
     }
@@ -182,6 +184,29 @@ class RsmiBackend final : public Backend {
             f.current < f.num_supported)
             s.mclk_mhz = f.frequency[f.current] / 1e6;
 
+        // RAS ECC totals over the blocks that exist on MI3xx
+        {
+            static const rsmi_gpu_block_t blocks[] = {
+                RSMI_GPU_BLOCK_UMC, RSMI_GPU_BLOCK_SDMA, RSMI_GPU_BLOCK_GFX,
+                RSMI_GPU_BLOCK_MMHUB, RSMI_GPU_BLOCK_XGMI_WAFL,
+                RSMI_GPU_BLOCK_PCIE_BIF};
+            double ce = 0, ue = 0;
+            bool any = false;
+            for (auto b : blocks) {
+                rsmi_error_count_t ec{};
+                if (api_.rsmi_dev_ecc_count_get(dev, b, &ec) ==
+                    RSMI_STATUS_SUCCESS) {
+                    ce += (double)ec.correctable_err;
+                    ue += (double)ec.uncorrectable_err;
+                    any = true;
+                }
+            }
+            if (any) {
+                s.ecc_correctable = ce;
+                s.ecc_uncorrectable = ue;
+            }
+        }
+
         rsmi_gpu_metrics_t gm;
         std::memset(&gm, 0, sizeof(gm));
         if (api_.rsmi_dev_gpu_metrics_info_get(dev, &gm) == RSMI_STATUS_SUCCESS) {
@@ -218,6 +243,16 @@ class RsmiBackend final : public Backend {
                 s.pcie_tx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
                 s.pcie_rx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
             }
+
+            if (gm.accumulation_counter && gm.accumulation_counter != ~0ull) {
+                s.accumulation_counter = (double)gm.accumulation_counter;
+                if (gm.ppt_residency_acc != ~0ull)
+                    s.ppt_residency_acc = (double)gm.ppt_residency_acc;
+                if (gm.socket_thm_residency_acc != ~0ull)
+                    s.thm_residency_acc = (double)gm.socket_thm_residency_acc;
+            }
+            if (gm.pcie_replay_count_acc != ~0ull)
+                s.pcie_replay_count = (double)gm.pcie_replay_count_acc;
 
             if (u16ok(gm.xgmi_link_width)) s.xgmi_link_width = gm.xgmi_link_width;
             if (u16ok(gm.xgmi_link_speed)) s.xgmi_link_speed_gbps = gm.xgmi_link_speed;

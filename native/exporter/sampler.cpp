@@ -64,6 +64,21 @@ void Sampler::sample_once()
                     (cur.gfx_activity_acc - prev.gfx_activity_acc) / dt_ms * 100.0;
                 if (pct <= 100.0) d.busy_windowed_pct = pct;
             }
+            // PVIOL/TVIOL per the gpu_metrics v1.6 formula
+            if (cur.accumulation_counter > 0 && prev.accumulation_counter > 0 &&
+                cur.accumulation_counter > prev.accumulation_counter) {
+                double dacc = cur.accumulation_counter - prev.accumulation_counter;
+                if (cur.ppt_residency_acc >= prev.ppt_residency_acc &&
+                    prev.ppt_residency_acc >= 0)
+                    d.pviol_pct =
+                        (cur.ppt_residency_acc - prev.ppt_residency_acc) * 100.0 /
+                        dacc;
+                if (cur.thm_residency_acc >= prev.thm_residency_acc &&
+                    prev.thm_residency_acc >= 0)
+                    d.tviol_pct =
+                        (cur.thm_residency_acc - prev.thm_residency_acc) * 100.0 /
+                        dacc;
+            }
             if (cur.mem_activity_acc >= 0 && prev.mem_activity_acc >= 0 &&
                 cur.mem_activity_acc >= prev.mem_activity_acc) {
                 double pct =

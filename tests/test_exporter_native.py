@@ -43,6 +43,21 @@ class TestMetricsContract:
         ]:
             assert fam in names, f"missing {fam}"
 
+    def test_ras_and_violation_families(self):
+        """ECC counters, PCIe replay, and windowed PVIOL/TVIOL (the mock
+        backend's residency accumulators advance at 10%/2% of wall)."""
+        with ExporterProcess(mock_devices=2, interval_ms=50) as exp:
+            time.sleep(0.3)  # windowed metrics need 2 samples
+            samples = parse_prometheus_text(exp.scrape())
+        by = {}
+        for s in samples:
+            by.setdefault(s.name, {})[s.labels.get("gpu")] = s.value
+        assert by["dcgm_ecc_sbe_aggregate_total"] == {"0": 0.0, "1": 1.0}
+        assert by["dcgm_ecc_dbe_aggregate_total"]["0"] == 0.0
+        assert by["dcgm_pcie_replay_counter"]["0"] == 3.0
+        assert by["dcgm_power_violation"]["0"] == pytest.approx(10.0, abs=1.5)
+        assert by["dcgm_thermal_violation"]["0"] == pytest.approx(2.0, abs=1.0)
+
     def test_amd_native_families(self):
         with ExporterProcess(mock_devices=1, interval_ms=50) as exp:
             time.sleep(0.3)  # need 2 samples for windowed rates
