@@ -19,6 +19,7 @@
 #include <memory>
 #include <mutex>
 #include <thread>
+#include <unistd.h>
 
 using namespace mi355x;
 
@@ -86,6 +87,10 @@ int main(int argc, char** argv)
     }
 
     RenderOptions ropt;
+    {
+        char host[256] = {0};
+        if (gethostname(host, sizeof(host) - 1) == 0) ropt.hostname = host;
+    }
     ropt.kubernetes = cfg.kubernetes;
     ropt.gpu_id_type = cfg.gpu_id_type;
     ropt.metric_set = cfg.metric_set;

@@ -50,6 +50,8 @@ class Writer {
         out_ << cur_family_ << "{gpu=\"" << d.info.index << "\",uuid=\""
              << esc(d.info.uuid) << "\",device=\"" << esc(d.info.drm_render)
              << "\",modelName=\"" << esc(d.info.name) << "\"";
+        if (!opt_.hostname.empty())
+            out_ << ",Hostname=\"" << esc(opt_.hostname) << "\"";
         if (opt_.kubernetes) {
             auto it = attr.find(attribution_key(d.info, opt_.gpu_id_type));
             if (it != attr.end()) {

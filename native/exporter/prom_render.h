@@ -37,6 +37,7 @@ using AttributionMap = std::map<std::string, PodAttribution>;
 
 struct RenderOptions {
     bool kubernetes = false;
+    std::string hostname;  // dcgm-exporter's Hostname label (empty = omit)
     std::string gpu_id_type = "device-name";
     // empty set = all metrics; otherwise only families named here (the
     // reference's `-f` metric-set file, one name per line, '#' comments).
