@@ -29,7 +29,9 @@ def amd_smi_busy():
         try:
             out = subprocess.run(args, capture_output=True, timeout=15)
             data = json.loads(out.stdout.decode())
-            # shape: [{"gpu":0, "usage": {"gfx_activity": {"value":N,...}}}]
+            # {"gpu_data": [{"gpu":0,"usage":{"gfx_activity":{"value":N}}}]}
+            if isinstance(data, dict):
+                data = data.get("gpu_data", [])
             if isinstance(data, list) and data:
                 usage = data[0].get("usage", {})
                 g = usage.get("gfx_activity")
