@@ -33,6 +33,12 @@ class HpaSpec:
     target_value: float = 5.0      # reference targetValue (cuda-test-hpa.yaml:21)
     tolerance: float = 0.1         # upstream default
     downscale_stabilization_s: float = 300.0  # upstream default
+    # v2 behavior.scaleUp policy (deploy/cuda-test-hpa.yaml's fix for the
+    # reference's documented overshoot, README.md:123): at most
+    # `scale_up_pods` added per `scale_up_period_s`. 0 = unlimited
+    # (v2beta1 / reference behavior).
+    scale_up_pods: int = 0
+    scale_up_period_s: float = 15.0
 
 
 @dataclass
@@ -125,6 +131,14 @@ def _stabilize(spec: HpaSpec, state: HpaState, desired: int, now_s: float) -> in
 
     if desired >= state.current_replicas:
         new = desired
+        if spec.scale_up_pods > 0 and new > state.current_replicas:
+            # v2 scaleUp policy: cap growth to scale_up_pods per period,
+            # measured against the replica count at the period's start
+            base_t, base_r = getattr(state, "scaleup_window", (None, None))
+            if base_t is None or now_s - base_t >= spec.scale_up_period_s:
+                base_t, base_r = now_s, state.current_replicas
+            new = min(new, base_r + spec.scale_up_pods)
+            state.scaleup_window = (base_t, base_r)
     else:
         new = max(d for (_, d) in state.recommendations)
         new = max(new, spec.min_replicas)

@@ -95,6 +95,25 @@ class TestReconcile:
         assert reconcile_multi(s, st, metrics, {"a": 0.0}, now_s=10.0) == 5
         assert reconcile_multi(s, st, metrics, {"a": 0.0}, now_s=61.0) == 1
 
+    def test_v2_scaleup_policy_prevents_overshoot(self):
+        """deploy/cuda-test-hpa.yaml's behavior block: 1 pod / 15 s. The
+        reference (v2beta1) jumps straight to maxReplicas (README.md:123);
+        the policy turns that into a stepped ramp."""
+        s = spec(max_replicas=8, scale_up_pods=1, scale_up_period_s=15.0)
+        st = HpaState(current_replicas=1)
+        counts = []
+        for i in range(8):
+            counts.append(reconcile(s, st, 40.0, now_s=i * 15.0))
+        assert counts == [2, 3, 4, 5, 6, 7, 8, 8]  # one pod per sync period
+
+    def test_v2_scaleup_policy_within_period(self):
+        s = spec(max_replicas=8, scale_up_pods=2, scale_up_period_s=15.0)
+        st = HpaState(current_replicas=1)
+        # two syncs inside one period add at most 2 pods total
+        assert reconcile(s, st, 40.0, now_s=0.0) == 3
+        assert reconcile(s, st, 40.0, now_s=5.0) == 3
+        assert reconcile(s, st, 40.0, now_s=15.0) == 5
+
     def test_scale_up_curve_to_eight(self):
         # config 4: 1 -> 8 replica scale-up under sustained high load
         s = spec(max_replicas=8)

@@ -39,3 +39,17 @@ def test_moderate_load_partial_scale():
     r = run("native", max_replicas=8, per_replica_busy=5.2, target=5.0,
             duration_s=60.0, pod_start_s=10.0)
     assert r["peak_desired"] == 1
+
+
+@needs_bin
+def test_native_v2_no_overshoot():
+    """The v2 behavior block removes the overshoot: replicas ramp one pod
+    per sync instead of the desired count jumping to max at once."""
+    from tools.scaleup_experiment import run
+
+    r = run("native-v2", max_replicas=8, per_replica_busy=40.0, target=5.0,
+            duration_s=200.0, pod_start_s=10.0)
+    assert r["overshoot"] is False
+    assert r["time_to_max_replicas_s"] is not None
+    desired_seq = [c["desired"] for c in r["curve"]]
+    assert max(desired_seq[:2]) <= 3  # stepped, not pinned at 8 immediately

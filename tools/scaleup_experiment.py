@@ -47,7 +47,12 @@ def run(profile: str, max_replicas: int, per_replica_busy: float,
     cadence = {
         "reference": dict(tick_s=10.0, rule_s=30.0, hpa_s=15.0),
         "native": dict(tick_s=1.0, rule_s=1.0, hpa_s=15.0),
+        # native cadences + the v2 behavior block from deploy/cuda-test-hpa
+        # (1 pod / 15 s): the stepped ramp that removes the reference's
+        # documented overshoot (README.md:123)
+        "native-v2": dict(tick_s=1.0, rule_s=1.0, hpa_s=15.0),
     }[profile]
+    scale_up_pods = 1 if profile == "native-v2" else 0
 
     with tempfile.TemporaryDirectory() as td:
         busy_file = Path(td) / "busy"
@@ -78,7 +83,8 @@ def run(profile: str, max_replicas: int, per_replica_busy: float,
             loop = ControlLoop(
                 scraper,
                 hpa_spec=HpaSpec(min_replicas=1, max_replicas=max_replicas,
-                                 target_value=target),
+                                 target_value=target,
+                                 scale_up_pods=scale_up_pods),
                 extra_samples=lambda: synth_pod_labels(
                     [f"cuda-test-{i}" for i in range(state["replicas"])]),
             )
@@ -135,7 +141,7 @@ def run(profile: str, max_replicas: int, per_replica_busy: float,
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--profile", choices=["reference", "native"],
+    ap.add_argument("--profile", choices=["reference", "native", "native-v2"],
                     default="native")
     ap.add_argument("--max-replicas", type=int, default=8)
     ap.add_argument("--busy", type=float, default=40.0,
