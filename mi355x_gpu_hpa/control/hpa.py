@@ -46,6 +46,8 @@ class HpaState:
     current_replicas: int = 1
     # (timestamp_s, desired) recommendations within the stabilization window
     recommendations: List[Tuple[float, int]] = field(default_factory=list)
+    # (period_start_s, replicas_at_period_start) for the scaleUp policy
+    scaleup_window: Tuple[Optional[float], Optional[int]] = (None, None)
 
 
 def desired_replicas(
@@ -134,7 +136,7 @@ def _stabilize(spec: HpaSpec, state: HpaState, desired: int, now_s: float) -> in
         if spec.scale_up_pods > 0 and new > state.current_replicas:
             # v2 scaleUp policy: cap growth to scale_up_pods per period,
             # measured against the replica count at the period's start
-            base_t, base_r = getattr(state, "scaleup_window", (None, None))
+            base_t, base_r = state.scaleup_window
             if base_t is None or now_s - base_t >= spec.scale_up_period_s:
                 base_t, base_r = now_s, state.current_replicas
             new = min(new, base_r + spec.scale_up_pods)
