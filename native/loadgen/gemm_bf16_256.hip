@@ -554,3 +554,178 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d8(
 {
     gemm_bf16_tn_256_impl<6, 0>(A, Bt, C, M, N, K, tiles_per_cta);
 }
+
+// ---------------------------------------------------------------------------
+// w32: the d6 schedule on v_mfma_f32_32x32x16_bf16 tiles.
+//
+// The 32x32 shape's measured ceiling is ~15% above 16x16's (2382 vs 2075
+// TF µbench) and each phase issues 8 long MFMAs instead of 16 short ones —
+// half the issue slots for the same FLOPs. Per-wave tile stays 128x64:
+// 4 m-frags x 2 n-frags of 32x32, acc 8 x 16 f32 = 128 regs (unchanged).
+// Fragment maps (CK/ISA): A: lane l -> A[i=l%32][k=8*(l/32)+j];
+// C/D: reg r -> row (r&3)+8*(r>>2)+4*(lane>>5), col lane&31.
+// Staging, liveness proofs, drains, swizzle and raster are d6's verbatim
+// (B is still read only at q0; A-halves live through q3).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_w32(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;
+    const int wc = w & 3;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    // 32x32 fragment read: row_in_half = frag_row + (lane&31),
+    // k = ks16*16 + (lane>>5)*8  -> byte off swizzled
+    auto frag_off32 = [&](int row_in_half, int ks16) {
+        return swz256(row_in_half * 128 + ks16 * 32 + ((lane >> 5) * 16));
+    };
+
+    const bool super4 = (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x16 acc[4][2];
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+#pragma unroll
+                for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        stage(1, 2, 1);
+        stage(1, 3, 1);
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[4]; // current m-frag, 4 k-steps of 16
+        bf16x8 bfrag[2][4];
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned short* la = &lds[(buf * 4 + wr) * HALF_HW];
+            const unsigned short* lb = &lds[(buf * 4 + 2 + (wc >> 1)) * HALF_HW];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int arow = q * 32 + (lane & 31);
+#pragma unroll
+                for (int ks = 0; ks < 4; ++ks)
+                    afrag[ks] = *(const bf16x8*)((const char*)la +
+                                                 frag_off32(arow, ks));
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 2; ++n) {
+                        const int bcol = bcol0 + n * 32 + (lane & 31);
+#pragma unroll
+                        for (int ks = 0; ks < 4; ++ks)
+                            bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                            frag_off32(bcol, ks));
+                    }
+                }
+
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 2, 2, buf);
+                } else if (q == 2) {
+                    stage(kt + 2, 3, buf);
+                }
+
+                __builtin_amdgcn_s_barrier();
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int n = 0; n < 2; ++n)
+#pragma unroll
+                    for (int ks = 0; ks < 4; ++ks)
+                        acc[q][n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                            afrag[ks], bfrag[n][ks], acc[q][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+
+                if (q == 3)
+                    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+                __builtin_amdgcn_s_barrier();
+            }
+        }
+
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+#pragma unroll
+            for (int j = 0; j < 2; ++j) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const long row = row0 + wr * 128 + i * 32 + (r & 3) +
+                                     8 * (r >> 2) + 4 * (lane >> 5);
+                    const long col = col0 + wc * 64 + j * 32 + (lane & 31);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
