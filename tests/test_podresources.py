@@ -217,9 +217,22 @@ class TestExporterK8sMode:
                     )
                 return None
 
+        state = {"response": response}
+
         sock = tmp_path / "kubelet.sock"
         server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
-        server.add_generic_rpc_handlers((Handler(),))
+
+        class MutableHandler(grpc.GenericRpcHandler):
+            def service(self, hcd):
+                if hcd.method == "/v1.PodResourcesLister/List":
+                    return grpc.unary_unary_rpc_method_handler(
+                        lambda req, ctx: state["response"],
+                        request_deserializer=None,
+                        response_serializer=None,
+                    )
+                return None
+
+        server.add_generic_rpc_handlers((MutableHandler(),))
         server.add_insecure_port(f"unix:{sock}")
         server.start()
         try:
@@ -235,14 +248,34 @@ class TestExporterK8sMode:
                                   if s.name == "dcgm_gpu_utilization"
                                   and s.labels.get("pod") == "cuda-test-xyz"]
                     _time.sleep(0.1)
-            assert attributed, "no pod-attributed series within 5s"
-            s = attributed[0]
-            assert s.labels["namespace"] == "default"
-            assert s.labels["container"] == "main"
-            assert s.labels["gpu"] == "0"
-            # device 1 (renderD129) is unallocated: no pod label
-            samples_by_gpu = {x.labels["gpu"]: x for x in samples
+                assert attributed, "no pod-attributed series within 5s"
+                s = attributed[0]
+                assert s.labels["namespace"] == "default"
+                assert s.labels["container"] == "main"
+                assert s.labels["gpu"] == "0"
+                # device 1 (renderD129) is unallocated: no pod label
+                samples_by_gpu = {x.labels["gpu"]: x for x in samples
+                                  if x.name == "dcgm_gpu_utilization"}
+                assert "pod" not in samples_by_gpu["1"].labels
+
+                # --- pod churn: the old pod dies, a new one gets the GPU;
+                # the attribution cache must invalidate (SURVEY.md §7
+                # "cache invalidation as pods churn") ---
+                state["response"] = list_response([
+                    pod("cuda-test-new", "default", [
+                        container("main", [container_devices(
+                            "amd.com/gpu", ["renderD129"])]),
+                    ]),
+                ])
+                deadline = _time.monotonic() + 5
+                churned = False
+                while _time.monotonic() < deadline and not churned:
+                    samples = parse_prometheus_text(exp.scrape())
+                    by_gpu = {x.labels["gpu"]: x for x in samples
                               if x.name == "dcgm_gpu_utilization"}
-            assert "pod" not in samples_by_gpu["1"].labels
+                    churned = (by_gpu["1"].labels.get("pod") == "cuda-test-new"
+                               and "pod" not in by_gpu["0"].labels)
+                    _time.sleep(0.1)
+                assert churned, "attribution did not follow the pod churn"
         finally:
             server.stop(0)
