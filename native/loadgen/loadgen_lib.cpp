@@ -227,11 +227,23 @@ static bool gemm_dims_ok(int m, int n, int k, int variant)
 // Timed GEMM: `iters` back-to-back launches after `warmup` untimed ones.
 // *ms_out = mean ms per GEMM, *tflops_out = 2*M*N*K / time.
 // variant: 1 = st_16x32-swizzled LDS (default), 0 = linear LDS (A/B ref).
+// 256-tile kernels need >=128 CTAs to fill the 256-CU chip; below that the
+// 128-tile kernel wins (4x the CTAs). Perf entry points auto-select; the
+// verify entry points do NOT (tests must exercise the exact kernel asked).
+static int gemm_effective_variant(int m, int n, int variant)
+{
+    if (variant >= 2 && (m / 256) * (n / 256) < 128 && m % 128 == 0 &&
+        n % 128 == 0)
+        return 1;
+    return variant;
+}
+
 int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
                                int iters, int variant, double* ms_out,
                                double* tflops_out)
 {
     if (!gemm_dims_ok(m, n, k, variant)) return -1;
+    variant = gemm_effective_variant(m, n, variant);
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
@@ -297,13 +309,14 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
     if (n <= 0) n = 4096;
     if (k <= 0) k = 4096;
     if (period_ms <= 0) period_ms = 100.0;
+    const int burn_variant = gemm_effective_variant(m, n, 2);
     if (target_util_pct < 0) target_util_pct = 0;
     if (target_util_pct > 100) target_util_pct = 100;
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
     // one calibration launch so the first period isn't all compile/warmup
-    gemm_launch(g, 0);
+    gemm_launch(g, 0, burn_variant);
     LG_CHECK(hipDeviceSynchronize());
 
     double t_end = now_ms() + seconds * 1e3;
@@ -312,7 +325,7 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
         double period_start = now_ms();
         double busy_until = period_start + period_ms * target_util_pct / 100.0;
         while (now_ms() < busy_until) {
-            gemm_launch(g, 0);
+            gemm_launch(g, 0, burn_variant);
             LG_CHECK(hipDeviceSynchronize());
         }
         double rest = period_start + period_ms - now_ms();
