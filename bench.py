@@ -253,6 +253,10 @@ def main():
     if exporter:
         final_replicas = loop.hpa_state.current_replicas if loop else None
         exporter.__exit__(None, None, None)
+        try:
+            os.unlink(mf.name)
+        except OSError:
+            pass
 
     if rank == 0:
         latencies.sort()

@@ -50,8 +50,11 @@ static __device__ __forceinline__ int swz256(int byte_off)
     return byte_off ^ (((byte_off >> 9) & 1) << 5);
 }
 
-// DEPTH: 2 = cross-boundary B0 prefetch (one half in flight across the
-// K-tile boundary); 1 = stage all of kt+1 during kt, full drain per K-tile.
+// DEPTH selects the measured schedule variants (see the header comment and
+// profiles/gemm_bf16_256_ladder.md): 1 = depth-1 full drain; 2 = one B0
+// half in flight; 4 = quadrant+all-B-held (spills); 5 = latency-balanced 2;
+// 6 = PRODUCT (m-quarter phases, all-B-held, B two tiles ahead).
+// RASTER = 1 adds the 4x4 super-tile rasterization.
 template <int DEPTH, int RASTER = 0>
 __device__ __forceinline__ void gemm_bf16_tn_256_impl(
     const unsigned short* __restrict__ A,  // [M][K] bf16

@@ -83,7 +83,9 @@ int main(int argc, char** argv)
             std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
-        const char* vname = variant == 0 ? "_linear" : variant == 2 ? "_256" : "";
+        char vname[16] = "";
+        if (variant == 0) std::snprintf(vname, sizeof(vname), "_linear");
+        else if (variant >= 2) std::snprintf(vname, sizeof(vname), "_v%d", variant);
         std::printf("gemm_bf16%s %dx%dx%d ms=%.3f tflops=%.1f\n",
                     vname, m, n, k, ms, tf);
         return 0;
