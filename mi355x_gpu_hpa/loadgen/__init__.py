@@ -78,6 +78,10 @@ def _load():
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_double, ctypes.c_void_p,
         ]
+        lib.lg_bw_burn.argtypes = [
+            ctypes.c_int, ctypes.c_double, ctypes.c_double, ctypes.c_double,
+            ctypes.c_double, ctypes.c_void_p, ctypes.POINTER(ctypes.c_double),
+        ]
         _lib = lib
     return _lib
 
@@ -145,3 +149,13 @@ def gemm_burn(target_util_pct: float, seconds: float, device: int = 0,
     """Duty-cycled GEMM load at a target GPU-busy percentage."""
     _check(_load().lg_gemm_burn(device, target_util_pct, seconds,
                                 m, n, k, period_ms, None))
+
+
+def bw_burn(target_util_pct: float, seconds: float, device: int = 0,
+            gb: float = 6.0, period_ms: float = 100.0) -> float:
+    """Duty-cycled streaming-triad HBM load (the bandwidth axis of the
+    multi-metric HPA). Returns achieved GB/s during the busy bursts."""
+    gbps = ctypes.c_double()
+    _check(_load().lg_bw_burn(device, target_util_pct, seconds, gb,
+                              period_ms, None, ctypes.byref(gbps)))
+    return gbps.value

@@ -27,6 +27,8 @@
 
 extern "C" __global__ void vector_add_f32(const float*, const float*, float*, int);
 extern "C" __global__ void vector_add_f32x4(const float4*, const float4*, float4*, int);
+extern "C" __global__ void triad_f32x4(const float4*, const float4*, float4*, float,
+                                       long);
 extern "C" __global__ void gemm_bf16_tn(const unsigned short*, const unsigned short*,
                                         float*, int, int, int, int);
 extern "C" __global__ void gemm_bf16_tn_linear(const unsigned short*,
@@ -295,6 +297,52 @@ int lg_gemm_bf16_verify(int device, const float* a_h, const float* bt_h,
                         float* c_out, int m, int n, int k)
 {
     return lg_gemm_bf16_verify_variant(device, a_h, bt_h, c_out, m, n, k, 1);
+}
+
+// Streaming-triad bandwidth burn: duty-cycled HBM traffic at
+// `target_util_pct` of wall time; `gb` working set (3 buffers summing to
+// ~gb GiB). Returns achieved GB/s over the busy bursts in *gbps_out.
+// The bandwidth-axis load for the multi-metric HPA (config 5).
+int lg_bw_burn(int device, double target_util_pct, double seconds, double gb,
+               double period_ms, volatile int* stop_flag, double* gbps_out)
+{
+    if (gb <= 0) gb = 6.0;
+    if (period_ms <= 0) period_ms = 100.0;
+    if (target_util_pct < 0) target_util_pct = 0;
+    if (target_util_pct > 100) target_util_pct = 100;
+    LG_CHECK(hipSetDevice(device));
+    long n4 = (long)(gb * (1ull << 30) / 3.0 / 16.0);
+    float4 *a, *b, *c;
+    LG_CHECK(hipMalloc(&a, n4 * 16));
+    LG_CHECK(hipMalloc(&b, n4 * 16));
+    LG_CHECK(hipMalloc(&c, n4 * 16));
+    LG_CHECK(hipMemset(a, 0x3f, n4 * 16));
+    LG_CHECK(hipMemset(b, 0x3e, n4 * 16));
+    int blocks = 8192;
+    double busy_ms_total = 0, bytes_total = 0;
+    double t_end = now_ms() + seconds * 1e3;
+    while (now_ms() < t_end) {
+        if (stop_flag && *stop_flag) break;
+        double period_start = now_ms();
+        double busy_until = period_start + period_ms * target_util_pct / 100.0;
+        while (now_ms() < busy_until) {
+            double t0 = now_ms();
+            hipLaunchKernelGGL(triad_f32x4, dim3(blocks), dim3(256), 0, 0, a, b,
+                               c, 1.5f, n4);
+            LG_CHECK(hipDeviceSynchronize());
+            busy_ms_total += now_ms() - t0;
+            bytes_total += 3.0 * n4 * 16;
+        }
+        double rest = period_start + period_ms - now_ms();
+        if (rest > 0)
+            std::this_thread::sleep_for(
+                std::chrono::duration<double, std::milli>(rest));
+    }
+    if (gbps_out)
+        *gbps_out = busy_ms_total > 0 ? bytes_total / (busy_ms_total * 1e-3) / 1e9
+                                      : 0;
+    (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
+    return 0;
 }
 
 // Duty-cycled GEMM burn: aim at `target_util_pct` GPU-busy for `seconds`.

@@ -29,6 +29,8 @@ int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
                                double* tflops_out);
 int lg_gemm_burn(int device, double target_util_pct, double seconds,
                  int m, int n, int k, double period_ms, volatile int* stop_flag);
+int lg_bw_burn(int device, double target_util_pct, double seconds, double gb,
+               double period_ms, volatile int* stop_flag, double* gbps_out);
 }
 
 static double argd(int argc, char** argv, const char* flag, double dflt)
@@ -42,7 +44,7 @@ int main(int argc, char** argv)
 {
     if (argc < 2) {
         std::fprintf(stderr,
-                     "usage: mi355x-loadgen {vectoradd|gemm|burn|devices} [flags]\n");
+                     "usage: mi355x-loadgen {vectoradd|gemm|burn|bwburn|devices} [flags]\n");
         return 1;
     }
     std::string mode = argv[1];
@@ -101,6 +103,20 @@ int main(int argc, char** argv)
             std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
+        return 0;
+    }
+    if (mode == "bwburn") {
+        double util = argd(argc, argv, "--util", 80.0);
+        double seconds = argd(argc, argv, "--seconds", 60.0);
+        double gb = argd(argc, argv, "--gb", 6.0);
+        double period = argd(argc, argv, "--period-ms", 100.0);
+        double gbps = 0;
+        if (lg_bw_burn(device, util, seconds, gb, period, nullptr, &gbps)) {
+            std::fprintf(stderr, "error: %s\n", lg_last_error());
+            return 2;
+        }
+        std::printf("bwburn util=%.0f%% achieved=%.0f GB/s during bursts\n",
+                    util, gbps);
         return 0;
     }
     std::fprintf(stderr, "unknown mode '%s'\n", mode.c_str());

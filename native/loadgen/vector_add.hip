@@ -50,3 +50,29 @@ extern "C" __global__ void __launch_bounds__(256) vector_add_f32x4(
         c[i] = vc;
     }
 }
+
+// Streaming triad: c[i] = a[i] + s*b[i], float4-vectorized, grid-stride.
+// The BANDWIDTH-bound load (3 x 16 B per lane-iteration, ~6 TB/s at full
+// duty): drives the HBM3E/UMC-activity metric family independently of
+// busy% so the multi-metric HPA (BASELINE config 5) can be exercised on
+// its bandwidth axis — the GEMM load is compute-heavy, this one is pure
+// memory traffic.
+extern "C" __global__ void __launch_bounds__(256) triad_f32x4(
+    const float4* __restrict__ a,
+    const float4* __restrict__ b,
+    float4* __restrict__ c,
+    float s, long n4)
+{
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (; i < n4; i += stride) {
+        float4 va = a[i];
+        float4 vb = b[i];
+        float4 vc;
+        vc.x = va.x + s * vb.x;
+        vc.y = va.y + s * vb.y;
+        vc.z = va.z + s * vb.z;
+        vc.w = va.w + s * vb.w;
+        c[i] = vc;
+    }
+}

@@ -129,3 +129,11 @@ def test_gemm_bench_sane(gpu):
     # MFMA path must be in play: even an untuned MFMA GEMM clears 100 TF/s;
     # a VALU/eager fallback cannot.
     assert tf > 100, f"bf16 GEMM at {tf:.0f} TF/s — MFMA path not engaged?"
+
+
+def test_bw_burn_hits_hbm(gpu):
+    """The streaming-triad load must push real HBM bandwidth (config 5's
+    bandwidth axis): >3 TB/s during bursts on MI355X (8 TB/s peak)."""
+    lg = _loadgen()
+    gbps = lg.bw_burn(100.0, 4.0, gb=6.0)
+    assert gbps > 3000, f"triad only {gbps:.0f} GB/s"
