@@ -55,7 +55,7 @@ static __device__ __forceinline__ int swz256(int byte_off)
 // half in flight; 4 = quadrant+all-B-held (spills); 5 = latency-balanced 2;
 // 6 = PRODUCT (m-quarter phases, all-B-held, B two tiles ahead).
 // RASTER = 1 adds the 4x4 super-tile rasterization.
-template <int DEPTH, int RASTER = 0>
+template <int DEPTH, int RASTER = 0, int SOFT_LGKM = 0>
 __device__ __forceinline__ void gemm_bf16_tn_256_impl(
     const unsigned short* __restrict__ A,  // [M][K] bf16
     const unsigned short* __restrict__ Bt, // [N][K] bf16
@@ -258,7 +258,14 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
                 else stage(kt + 1, 2, buf ^ 1);
 
                 __builtin_amdgcn_s_barrier();
-                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                // SOFT_LGKM: rely on hipcc's counted lgkm waits between each
+                // ds_read and its consuming MFMA instead of a full drain —
+                // the first MFMA can issue ~50 cyc earlier per phase. All
+                // reads still retire before the phase-closing barrier (their
+                // consumers issue before it), so the staging-deadness proofs
+                // hold unchanged.
+                if (!SOFT_LGKM)
+                    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
                 __builtin_amdgcn_s_setprio(1);
                 if (DEPTH == 6) {
@@ -731,4 +738,12 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_w32(
         }
         __syncthreads();
     }
+}
+
+// product schedule with compiler-counted lgkm waits (no post-barrier drain).
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_soft(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl<6, 1, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
