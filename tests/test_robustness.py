@@ -98,3 +98,47 @@ class TestHttpRobustness:
         for _ in range(100):
             self._raw(exporter.port, b"GET /healthz HTTP/1.1\r\n\r\n")
         assert "dcgm_gpu_utilization" in exporter.scrape()
+
+
+@needs_lib
+def test_fuzz_under_asan():
+    """Deeper fuzz against an ASAN+UBSAN build of the parser: any OOB read
+    or UB aborts the interpreter (detected as a crash here)."""
+    import shutil
+    import subprocess
+    from pathlib import Path
+
+    native = Path(__file__).resolve().parent.parent / "native"
+    if not shutil.which("g++"):
+        pytest.skip("no g++")
+    r = subprocess.run(["make", "-C", str(native), "asan-lib"],
+                       capture_output=True)
+    asan_lib = native / "build" / "libmi355x_sampler_asan.so"
+    if r.returncode != 0 or not asan_lib.exists():
+        pytest.skip("asan lib unavailable")
+    # run the fuzz loop in a subprocess so an ASAN abort fails cleanly
+    code = f"""
+import ctypes, random
+lib = ctypes.CDLL({str(asan_lib)!r})
+lib.mi355x_parse_list_response_json.argtypes = [
+    ctypes.c_char_p, ctypes.c_int, ctypes.c_char_p, ctypes.c_int]
+buf = ctypes.create_string_buffer(1 << 16)
+rng = random.Random(7)
+for _ in range(20000):
+    n = rng.randrange(0, 300)
+    data = bytes(rng.randrange(256) for _ in range(n))
+    rc = lib.mi355x_parse_list_response_json(data, n, buf, len(buf))
+    assert rc >= -1
+print("fuzz clean")
+"""
+    import os
+
+    asan_rt = subprocess.run(["g++", "-print-file-name=libasan.so"],
+                             capture_output=True).stdout.decode().strip()
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = asan_rt
+    env["ASAN_OPTIONS"] = "detect_leaks=0"  # python itself leaks by design
+    p = subprocess.run(["python3", "-c", code], capture_output=True,
+                       timeout=300, env=env)
+    assert p.returncode == 0, (p.stdout.decode()[-500:], p.stderr.decode()[-2000:])
+    assert b"fuzz clean" in p.stdout
