@@ -53,16 +53,18 @@ class HpaState:
 def desired_replicas(
     spec: HpaSpec, current_replicas: int, metric_value: Optional[float]
 ) -> int:
-    """Raw desired-replica computation (no stabilization)."""
+    """Raw desired-replica computation (no stabilization). The min/max
+    clamp applies even on the no-change paths (a missing metric or the
+    tolerance band): the controller always enforces the replica bounds."""
+    clamp = lambda r: max(spec.min_replicas, min(spec.max_replicas, r))  # noqa: E731
     if metric_value is None:
-        return current_replicas
+        return clamp(current_replicas)
     if spec.target_value <= 0:
         raise ValueError("target_value must be positive")
     ratio = metric_value / spec.target_value
     if abs(ratio - 1.0) <= spec.tolerance:
-        return current_replicas
-    desired = math.ceil(ratio * current_replicas)
-    return max(spec.min_replicas, min(spec.max_replicas, desired))
+        return clamp(current_replicas)
+    return clamp(math.ceil(ratio * current_replicas))
 
 
 @dataclass

@@ -1,0 +1,88 @@
+"""Property-based tests (hypothesis) for the control-plane math."""
+
+import math
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from mi355x_gpu_hpa.control import HpaSpec, Sample, desired_replicas, evaluate
+
+values = st.floats(min_value=0, max_value=1000, allow_nan=False,
+                   allow_infinity=False)
+
+
+class TestHpaProperties:
+    @given(st.lists(values, min_size=2, max_size=20),
+           st.integers(min_value=1, max_value=8))
+    @settings(max_examples=200, deadline=None)
+    def test_desired_monotone_in_metric(self, vals, replicas):
+        spec = HpaSpec(min_replicas=1, max_replicas=64, target_value=5.0)
+        got = [desired_replicas(spec, replicas, v) for v in sorted(vals)]
+        assert got == sorted(got)  # non-decreasing in the metric
+
+    @given(values, st.integers(min_value=1, max_value=64))
+    @settings(max_examples=200, deadline=None)
+    def test_desired_within_bounds(self, v, replicas):
+        spec = HpaSpec(min_replicas=2, max_replicas=16, target_value=5.0)
+        d = desired_replicas(spec, replicas, v)
+        assert 2 <= d <= 16
+
+    @given(st.floats(min_value=0.91, max_value=1.09),
+           st.integers(min_value=1, max_value=16))
+    @settings(max_examples=100, deadline=None)
+    def test_tolerance_band_no_change(self, ratio, replicas):
+        spec = HpaSpec(min_replicas=1, max_replicas=64, target_value=10.0)
+        assert desired_replicas(spec, replicas, 10.0 * ratio) == replicas
+
+
+class TestPromqlProperties:
+    @given(st.lists(st.tuples(st.sampled_from(["a", "b", "c"]), values),
+                    min_size=1, max_size=30))
+    @settings(max_examples=200, deadline=None)
+    def test_aggregations_match_model(self, pairs):
+        samples = [Sample("m", {"g": g, "i": str(i)}, v)
+                   for i, (g, v) in enumerate(pairs)]
+        vals = [v for _, v in pairs]
+        assert evaluate("sum(m)", samples)[0].value == sum(vals)
+        assert evaluate("max(m)", samples)[0].value == max(vals)
+        assert evaluate("min(m)", samples)[0].value == min(vals)
+        assert math.isclose(evaluate("avg(m)", samples)[0].value,
+                            sum(vals) / len(vals), rel_tol=1e-9)
+        # group sums partition the total
+        by_g = evaluate("sum by(g) (m)", samples)
+        assert math.isclose(sum(s.value for s in by_g), sum(vals),
+                            rel_tol=1e-9)
+
+    @given(st.lists(st.tuples(st.sampled_from(["p0", "p1", "p2"]), values),
+                    min_size=1, max_size=20))
+    @settings(max_examples=100, deadline=None)
+    def test_join_filters_to_labeled_pods(self, pairs):
+        samples = [Sample("util", {"pod": p, "i": str(i)}, v)
+                   for i, (p, v) in enumerate(pairs)]
+        samples.append(Sample("kube_pod_labels",
+                              {"pod": "p0", "label_app": "x"}, 1.0))
+        expr = ('avg(max by(pod) (util) * on(pod) group_left(label_app) '
+                'max by(pod, label_app) (kube_pod_labels{label_app="x"}))')
+        res = evaluate(expr, samples)
+        p0 = [v for p, v in pairs if p == "p0"]
+        if not p0:
+            assert res == []
+        else:
+            assert math.isclose(res[0].value, max(p0), rel_tol=1e-9)
+
+
+class TestRoundTrip:
+    @given(st.lists(values, min_size=1, max_size=8))
+    @settings(max_examples=50, deadline=None)
+    def test_render_parse_roundtrip(self, vals):
+        """Values written in the exporter's %.6g format parse back to within
+        6 significant digits."""
+        from mi355x_gpu_hpa.control import parse_prometheus_text
+
+        text = "\n".join(
+            f'dcgm_gpu_utilization{{gpu="{i}"}} {v:.6g}'
+            for i, v in enumerate(vals))
+        parsed = parse_prometheus_text(text)
+        assert len(parsed) == len(vals)
+        for s, v in zip(parsed, vals):
+            assert math.isclose(s.value, v, rel_tol=1e-5, abs_tol=1e-4)
