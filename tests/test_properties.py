@@ -86,3 +86,39 @@ class TestRoundTrip:
         assert len(parsed) == len(vals)
         for s, v in zip(parsed, vals):
             assert math.isclose(s.value, v, rel_tol=1e-5, abs_tol=1e-4)
+
+
+class TestNativeRendererProperties:
+    """Hypothesis through the C++ renderer (capi, mock backend): injected
+    busy% must come back exactly through render -> parse."""
+
+    @given(st.floats(min_value=0, max_value=100, allow_nan=False))
+    @settings(max_examples=50, deadline=None)
+    def test_injected_busy_roundtrip(self, busy):
+        import ctypes
+        import os
+
+        from mi355x_gpu_hpa import NATIVE_BUILD
+        from mi355x_gpu_hpa.control import parse_prometheus_text
+
+        lib_path = NATIVE_BUILD / "libmi355x_sampler.so"
+        if not lib_path.exists():
+            import pytest
+
+            pytest.skip("sampler lib not built")
+        lib = ctypes.CDLL(str(lib_path))
+        lib.mi355x_render_mock_metrics.argtypes = [
+            ctypes.c_int, ctypes.c_char_p, ctypes.c_char_p,
+            ctypes.c_char_p, ctypes.c_int]
+        os.environ["MI355X_MOCK_BUSY"] = f"{busy!r}"
+        try:
+            buf = ctypes.create_string_buffer(1 << 18)
+            rc = lib.mi355x_render_mock_metrics(2, b"", b"", buf, len(buf))
+            assert rc > 0
+            samples = parse_prometheus_text(buf.value.decode())
+            utils = [s for s in samples if s.name == "dcgm_gpu_utilization"]
+            assert len(utils) == 2
+            for s in utils:
+                assert math.isclose(s.value, busy, rel_tol=1e-5, abs_tol=1e-4)
+        finally:
+            del os.environ["MI355X_MOCK_BUSY"]
