@@ -148,3 +148,24 @@ class TestScrapeValues:
         relabels = job["relabel_configs"]
         node = next(r for r in relabels if r.get("target_label") == "node")
         assert node["source_labels"] == ["__meta_kubernetes_pod_node_name"]
+
+
+class TestKustomize:
+    def test_base_lists_existing_resources(self):
+        k = yaml.safe_load((DEPLOY / "kustomization.yaml").read_text())
+        for res in k["resources"]:
+            assert (DEPLOY / res).exists(), res
+        # every standalone top-level manifest is included
+        listed = set(k["resources"])
+        present = {p.name for p in DEPLOY.glob("*.yaml")
+                   if p.name not in ("kustomization.yaml",
+                                     "kube-prometheus-stack-values.yaml")}
+        assert listed == present
+
+    def test_overlay(self):
+        k = yaml.safe_load(
+            (DEPLOY / "multi-metric" / "kustomization.yaml").read_text())
+        assert "../" in k["resources"]
+        for res in k["resources"]:
+            if res != "../":
+                assert (DEPLOY / "multi-metric" / res).exists(), res
