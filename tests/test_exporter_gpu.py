@@ -114,6 +114,21 @@ def test_duty_cycle_tracks_target(gpu):
     assert 25 <= mean <= 75, f"mean busy {mean}% for 50% duty target ({vals})"
 
 
+def test_live_schema_superset_of_fixtures(gpu):
+    """The committed CPU fixtures must never claim families the live
+    exporter no longer serves (fixture drift guard)."""
+    from pathlib import Path
+
+    fixture = (Path(__file__).parent / "fixtures" /
+               "real_mi355x_idle.prom").read_text()
+    fixture_families = {s.name for s in parse_prometheus_text(fixture)}
+    with ExporterProcess(interval_ms=200) as exp:
+        time.sleep(0.5)
+        live_families = {s.name for s in parse_prometheus_text(exp.scrape())}
+    missing = fixture_families - live_families
+    assert not missing, f"live exporter dropped families: {missing}"
+
+
 def test_readyz_real_backend(gpu):
     import urllib.request
 
