@@ -138,6 +138,7 @@ def main():
     exporter = None
     loop = None
     util_err_pct = None
+    util_err_per_gpu = None
     n_exported = 0
     if rank == 0:
         # Narrow the exporter to the families the decision path consumes
@@ -250,6 +251,7 @@ def main():
         if per_gpu_errs:
             medians = {g: statistics.median(v) for g, v in per_gpu_errs.items()}
             util_err_pct = max(medians.values())
+            util_err_per_gpu = {str(g): round(v, 2) for g, v in medians.items()}
             log(f"util err vs rocm-smi (median of 5 paired reads per GPU): "
                 f"{medians} -> max {util_err_pct:.1f}%")
         xgmi_bps = [s.value for s in samples
@@ -295,6 +297,7 @@ def main():
                 "parallelism": f"replicas{n_gpus}",
                 "p99_ms": round(p99_ms, 3),
                 "util_err_vs_rocm_smi_pct": util_err_pct,
+                "util_err_per_gpu_pct": util_err_per_gpu,
                 "gpus_exported": n_exported,
                 "final_replicas": final_replicas,
                 "reference_cadence_s": 10.0,
