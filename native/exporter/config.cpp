@@ -18,6 +18,7 @@ const char* config_usage()
            "  --pod-resources-socket <path>\n"
            "  --mock <n>                  mock backend with n synthetic GPUs\n"
            "  --mock-busy-file <path>     scriptable busy%% for the mock backend\n"
+           "  -v, --version\n"
            "  -h, --help\n";
 }
 
@@ -94,6 +95,9 @@ bool parse_config(int argc, char** argv, Config* cfg, std::string* err)
         const char* v;
         if (a == "-h" || a == "--help") {
             cfg->show_help = true;
+            return true;
+        } else if (a == "-v" || a == "--version") {
+            cfg->show_version = true;
             return true;
         } else if (a == "-c") {
             if (!(v = need(i))) goto missing;

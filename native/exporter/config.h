@@ -37,6 +37,7 @@ struct Config {
     int mock_devices = 0;             // >0 => mock backend
     std::string mock_busy_file;
     bool show_help = false;
+    bool show_version = false;
 };
 
 // Returns false + err message on a bad flag. argv-style parsing.

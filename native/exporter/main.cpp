@@ -38,6 +38,10 @@ int main(int argc, char** argv)
         std::fputs(config_usage(), stdout);
         return 0;
     }
+    if (cfg.show_version) {
+        std::puts("mi355x-exporter 0.1.0 (gfx950; rocm_smi backend)");
+        return 0;
+    }
 
     std::unique_ptr<Backend> backend;
     if (cfg.mock_devices > 0) {

@@ -197,3 +197,34 @@ class TestConfig:
         finally:
             p.terminate()
             p.wait(timeout=5)
+
+
+@needs_bin
+class TestConfigEdgeCases:
+    def run(self, *args):
+        return subprocess.run([EXPORTER_BIN, *args], capture_output=True,
+                              timeout=15)
+
+    def test_version_flag(self):
+        r = self.run("--version")
+        assert r.returncode == 0
+        assert b"mi355x-exporter" in r.stdout
+
+    def test_bad_listen_spec(self):
+        r = self.run("--mock", "1", "-c", "50", "-l", "not a spec")
+        assert r.returncode == 2
+        assert b"bad listen spec" in r.stderr
+
+    def test_interval_floor(self):
+        r = self.run("--mock", "1", "-c", "1")
+        assert r.returncode == 2
+        assert b"below 10 ms" in r.stderr
+
+    def test_bad_gpu_id_type(self):
+        r = self.run("--mock", "1", "--kubernetes-gpu-id-type", "bogus")
+        assert r.returncode == 2
+
+    def test_missing_metric_file(self):
+        r = self.run("--mock", "1", "-f", "/nonexistent/metrics.csv")
+        assert r.returncode == 2
+        assert b"cannot open" in r.stderr
