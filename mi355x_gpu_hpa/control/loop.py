@@ -23,7 +23,7 @@ from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional
 
 from .hpa import HpaSpec, HpaState, reconcile
-from .promql import Sample, evaluate
+from .promql import PromQLError, Sample, evaluate
 from .scraper import Scraper
 
 # The reference recording rule, verbatim semantics
@@ -74,6 +74,9 @@ class ControlLoop:
         self.hpa_metric = hpa_metric
         self.extra_samples = extra_samples
         self.recorded_series: List[Sample] = []
+        # rule name -> first error message (a failing rule is reported once,
+        # like Prometheus's unhealthy-rule state)
+        self._rule_errors: Dict[str, str] = {}
 
     def step(self, now_s: Optional[float] = None) -> LoopResult:
         t0 = time.monotonic()
@@ -85,7 +88,16 @@ class ControlLoop:
         recorded: Dict[str, Optional[float]] = {}
         self.recorded_series = []
         for rule in self.rules:
-            vec = evaluate(rule.expr, samples)
+            try:
+                vec = evaluate(rule.expr, samples)
+            except PromQLError as e:
+                # Prometheus marks a failing rule as unhealthy and keeps
+                # evaluating the group; mirror that instead of crashing the
+                # loop on a bad custom rule.
+                if rule.record not in self._rule_errors:
+                    self._rule_errors[rule.record] = str(e)
+                recorded[rule.record] = None
+                continue
             if not vec:
                 recorded[rule.record] = None
                 continue

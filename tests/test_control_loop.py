@@ -107,3 +107,21 @@ def test_loop_without_exporter_is_robust():
     r = loop.step()
     assert r.metric_value is None
     assert r.replicas == 1
+
+
+def test_malformed_rule_does_not_crash_loop():
+    """Prometheus marks a failing rule unhealthy and keeps going; so do we."""
+    from mi355x_gpu_hpa.control import RecordingRule
+
+    scraper = Scraper([ScrapeTarget("http://127.0.0.1:1/metrics")],
+                      timeout_s=0.2)
+    loop = ControlLoop(scraper, rules=[
+        RecordingRule("bad_rule", "avg(((broken"),
+        RecordingRule("good_rule", "avg(kube_pod_labels)"),
+    ], extra_samples=lambda: synth_pod_labels(["p"]))
+    r = loop.step()
+    assert r.recorded["bad_rule"] is None
+    assert r.recorded["good_rule"] == 1.0
+    assert "bad_rule" in loop._rule_errors
+    # second step: still alive, error reported once
+    loop.step()
