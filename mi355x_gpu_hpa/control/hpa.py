@@ -133,18 +133,21 @@ def _stabilize(spec: HpaSpec, state: HpaState, desired: int, now_s: float) -> in
     cutoff = now_s - spec.downscale_stabilization_s
     state.recommendations = [(t, d) for (t, d) in state.recommendations if t >= cutoff]
 
-    if desired >= state.current_replicas:
-        new = desired
-        if spec.scale_up_pods > 0 and new > state.current_replicas:
-            # v2 scaleUp policy: cap growth to scale_up_pods per period,
-            # measured against the replica count at the period's start
-            base_t, base_r = state.scaleup_window
-            if base_t is None or now_s - base_t >= spec.scale_up_period_s:
-                base_t, base_r = now_s, state.current_replicas
-            new = min(new, base_r + spec.scale_up_pods)
-            state.scaleup_window = (base_t, base_r)
-    else:
-        new = max(d for (_, d) in state.recommendations)
-        new = max(new, spec.min_replicas)
+    # kube-controller-manager ordering: (1) stabilization — the highest
+    # recommendation within the window wins, which is what delays
+    # downscale; (2) behavior rate limits apply AFTER stabilization, so a
+    # replayed high recommendation that was rate-limited earlier cannot
+    # bypass the scaleUp policy on a later dip.
+    stabilized = max(d for (_, d) in state.recommendations)
+    new = stabilized
+    if spec.scale_up_pods > 0 and new > state.current_replicas:
+        # v2 scaleUp policy: cap growth to scale_up_pods per period,
+        # measured against the replica count at the period's start
+        base_t, base_r = state.scaleup_window
+        if base_t is None or now_s - base_t >= spec.scale_up_period_s:
+            base_t, base_r = now_s, state.current_replicas
+        new = min(new, base_r + spec.scale_up_pods)
+        state.scaleup_window = (base_t, base_r)
+    new = max(spec.min_replicas, min(spec.max_replicas, new))
     state.current_replicas = new
     return new

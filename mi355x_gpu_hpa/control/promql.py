@@ -159,32 +159,52 @@ class _Parser:
             raise PromQLError(f"trailing input at {self.peek()[1]!r}")
         return e
 
+    # Precedence follows Prometheus: * / bind tighter than + - (both
+    # levels left-associative), so e.g. `a + b * c` is `a + (b * c)`.
     def parse_binary(self) -> Expr:
+        return self.parse_additive()
+
+    def parse_additive(self) -> Expr:
+        lhs = self.parse_multiplicative()
+        while True:
+            t = self.peek()
+            if t is None or t[1] not in ("+", "-"):
+                return lhs
+            op = self.next()[1]
+            on, group_left = self.parse_match_modifiers()
+            rhs = self.parse_multiplicative()
+            lhs = BinOp(op, on, group_left, lhs, rhs)
+
+    def parse_multiplicative(self) -> Expr:
         lhs = self.parse_primary()
         while True:
             t = self.peek()
-            if t is None or t[1] not in ("*", "/", "+", "-"):
+            if t is None or t[1] not in ("*", "/"):
                 return lhs
             op = self.next()[1]
-            on = group_left = None
-            t = self.peek()
-            if t and t[0] == "id" and t[1] in ("on", "ignoring"):
-                kind = self.next()[1]
-                names = self.parse_name_list()
-                if kind == "on":
-                    on = names
-                else:
-                    raise PromQLError("ignoring() not supported; use on()")
-            t = self.peek()
-            if t and t[0] == "id" and t[1] in ("group_left", "group_right"):
-                kind = self.next()[1]
-                if kind == "group_right":
-                    raise PromQLError("group_right not supported")
-                group_left = []
-                if self.peek() and self.peek()[1] == "(":
-                    group_left = self.parse_name_list()
+            on, group_left = self.parse_match_modifiers()
             rhs = self.parse_primary()
             lhs = BinOp(op, on, group_left, lhs, rhs)
+
+    def parse_match_modifiers(self):
+        on = group_left = None
+        t = self.peek()
+        if t and t[0] == "id" and t[1] in ("on", "ignoring"):
+            kind = self.next()[1]
+            names = self.parse_name_list()
+            if kind == "on":
+                on = names
+            else:
+                raise PromQLError("ignoring() not supported; use on()")
+        t = self.peek()
+        if t and t[0] == "id" and t[1] in ("group_left", "group_right"):
+            kind = self.next()[1]
+            if kind == "group_right":
+                raise PromQLError("group_right not supported")
+            group_left = []
+            if self.peek() and self.peek()[1] == "(":
+                group_left = self.parse_name_list()
+        return on, group_left
 
     def parse_name_list(self) -> List[str]:
         self.expect("(")

@@ -53,11 +53,14 @@ struct Cursor {
         return true;
     }
 
-    // returns (ptr,len) for a length-delimited field
+    // returns (ptr,len) for a length-delimited field. The length is an
+    // unvalidated varint from the wire: compare LENGTHS, never `p + l`
+    // (a huge l would overflow the pointer arithmetic — UB — before the
+    // bound check could reject it).
     bool bytes(const uint8_t** data, size_t* len)
     {
         uint64_t l = varint();
-        if (!ok || p + l > end) {
+        if (!ok || l > (uint64_t)(end - p)) {
             ok = false;
             return false;
         }
@@ -71,17 +74,22 @@ struct Cursor {
     {
         switch (wire) {
             case 0: varint(); break;
-            case 1: p += 8; break;
+            case 1:
+                if (8 > (size_t)(end - p)) { ok = false; return; }
+                p += 8;
+                break;
             case 2: {
                 const uint8_t* d;
                 size_t l;
                 bytes(&d, &l);
                 break;
             }
-            case 5: p += 4; break;
+            case 5:
+                if (4 > (size_t)(end - p)) { ok = false; return; }
+                p += 4;
+                break;
             default: ok = false;
         }
-        if (p > end) ok = false;
     }
 };
 

@@ -114,6 +114,22 @@ class TestReconcile:
         assert reconcile(s, st, 40.0, now_s=5.0) == 3
         assert reconcile(s, st, 40.0, now_s=15.0) == 5
 
+    def test_stabilized_replay_cannot_bypass_scaleup_policy(self):
+        """behavior rate limits must apply AFTER
+        stabilization (kube-controller-manager ordering). An earlier
+        desired=8 that the scaleUp policy limited to 2 must not be
+        replayed uncapped by the downscale-stabilization max on a later
+        dip."""
+        s = spec(max_replicas=8, scale_up_pods=1, scale_up_period_s=15.0,
+                 downscale_stabilization_s=300.0)
+        st = HpaState(current_replicas=1)
+        # burst: desired=8, rate-limited to +1/period
+        assert reconcile(s, st, 40.0, now_s=0.0) == 2
+        # dip inside the stabilization window: the windowed max (8) is
+        # replayed by stabilization but must still be rate-limited
+        assert reconcile(s, st, 1.0, now_s=15.0) == 3
+        assert reconcile(s, st, 1.0, now_s=30.0) == 4
+
     def test_scale_up_curve_to_eight(self):
         # config 4: 1 -> 8 replica scale-up under sustained high load
         s = spec(max_replicas=8)

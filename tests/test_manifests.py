@@ -31,17 +31,27 @@ class TestExporterManifest:
         assert c["ports"][0]["containerPort"] == 9400
         assert self.svc["spec"]["ports"][0]["port"] == 9400
 
-    def test_no_privileged_no_nvidia(self):
+    def test_no_sysadmin_no_nvidia(self):
         # check live YAML content (comments cite the reference and may name
-        # NVIDIA paths; the actual spec must not)
+        # NVIDIA paths; the actual spec must not). Unlike the reference
+        # (dcgm-exporter.yaml:42-48) there is no SYS_ADMIN capability;
+        # `privileged: true` IS expected — runc's device cgroup blocks
+        # open() of hostPath char devices in non-privileged containers
+        # (ADVICE round 1, medium).
         text = "\n".join(line for line in
                          (DEPLOY / "mi355x-exporter.yaml").read_text().splitlines()
                          if not line.lstrip().startswith("#"))
         for line in text.splitlines():
             stripped = line.split("#")[0].lower()
-            assert "privileged" not in stripped
             assert "sys_admin" not in stripped
             assert "nvidia" not in stripped, line
+
+    def test_privileged_for_device_cgroup(self):
+        sc = self.ds["spec"]["template"]["spec"]["containers"][0]["securityContext"]
+        assert sc.get("privileged") is True
+        # and no added capabilities — privilege is device-cgroup passthrough
+        # only, not a capability grant
+        assert "capabilities" not in sc
 
     def test_amd_device_paths(self):
         vols = {v["name"]: v for v in self.ds["spec"]["template"]["spec"]["volumes"]}
@@ -148,6 +158,19 @@ class TestScrapeValues:
         relabels = job["relabel_configs"]
         node = next(r for r in relabels if r.get("target_label") == "node")
         assert node["source_labels"] == ["__meta_kubernetes_pod_node_name"]
+
+    def test_ksm_label_allowlist(self):
+        # kube-state-metrics >=2.0 exports no
+        # label_* labels unless allowlisted — without this the rule's
+        # `on(pod) group_left(label_app)` join is empty in a real cluster
+        # and the HPA reads <unknown>.
+        v = load_all(DEPLOY / "kube-prometheus-stack-values.yaml")[0]
+        args = v["kube-state-metrics"]["extraArgs"]
+        allow = next(a for a in args
+                     if a.startswith("--metric-labels-allowlist="))
+        spec = allow.split("=", 1)[1]
+        # must cover pods' `app` label, the join key of the recording rule
+        assert "pods=[" in spec and "app" in spec.split("pods=[", 1)[1]
 
 
 class TestKustomize:

@@ -141,6 +141,29 @@ class TestProtobufWalker:
         assert rc == -1
         assert buf.value.startswith(b"ERR")
 
+    def test_huge_varint_length_rejected(self):
+        # a near-2^64 length-delimited size used to
+        # overflow `p + l` pointer arithmetic before the bound check; the
+        # guard now compares lengths. Must reject, not crash or accept.
+        lib = _lib()
+        for huge in (2**64 - 1, 2**63, 2**32 + 7):
+            buf = ctypes.create_string_buffer(4096)
+            bad = _tag(1, 2) + _varint(huge) + b"x"
+            rc = lib.mi355x_parse_list_response_json(bad, len(bad),
+                                                     buf, len(buf))
+            assert rc == -1, huge
+
+    def test_truncated_fixed_width_rejected(self):
+        # wire types 1 (64-bit) and 5 (32-bit) with fewer payload bytes
+        # than the width must fail cleanly (skip() bound check)
+        lib = _lib()
+        for wire, payload in ((1, b"\x01\x02"), (5, b"\x01")):
+            buf = ctypes.create_string_buffer(4096)
+            bad = _tag(7, wire) + payload
+            rc = lib.mi355x_parse_list_response_json(bad, len(bad),
+                                                     buf, len(buf))
+            assert rc == -1, wire
+
 
 @needs_lib
 class TestAgainstRealGrpcServer:

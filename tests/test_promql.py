@@ -137,6 +137,21 @@ class TestPrimitives:
         samples = [Sample("l", {"pod": "p"}, 2.0), Sample("r", {"pod": "q"}, 3.0)]
         assert evaluate("l * on(pod) r", samples) == []
 
+    def test_operator_precedence(self):
+        """* and / bind tighter than + and -
+        (Prometheus precedence), so `a + b * c` is `a + (b * c)`."""
+        samples = [
+            Sample("a", {"k": "x"}, 2.0),
+            Sample("b", {"k": "x"}, 3.0),
+            Sample("c", {"k": "x"}, 4.0),
+        ]
+        assert evaluate("a + b * c", samples)[0].value == 14.0
+        assert evaluate("a - b / c", samples)[0].value == 2.0 - 3.0 / 4.0
+        assert evaluate("(a + b) * c", samples)[0].value == 20.0
+        # scalar forms too
+        assert evaluate("2 + 3 * 4", [])[0].value == 14.0
+        assert evaluate("a * 100 - b", samples)[0].value == 197.0
+
     def test_parse_errors(self):
         with pytest.raises(PromQLError):
             evaluate("avg(", [])
