@@ -1,3 +1,4 @@
+from .adapter import Adapter, AdapterError, DiscoveredMetric, discover
 from .hpa import (
     HpaSpec,
     HpaState,
@@ -19,6 +20,7 @@ from .promql import PromQLError, Sample, evaluate, evaluate_scalar
 from .scraper import Scraper, ScrapeTarget, parse_prometheus_text
 
 __all__ = [
+    "Adapter", "AdapterError", "DiscoveredMetric", "discover",
     "HpaSpec", "HpaState", "MetricTarget", "desired_replicas",
     "desired_replicas_multi", "reconcile", "reconcile_multi",
     "ControlLoop", "LoopResult", "RecordingRule", "synth_pod_labels",

@@ -22,6 +22,7 @@ import time
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional
 
+from .adapter import Adapter
 from .hpa import HpaSpec, HpaState, reconcile
 from .promql import PromQLError, Sample, evaluate
 from .scraper import Scraper
@@ -53,6 +54,7 @@ class LoopResult:
     recorded: Dict[str, Optional[float]]
     replicas: int
     metric_value: Optional[float]
+    adapter_s: float = 0.0
 
 
 class ControlLoop:
@@ -63,6 +65,8 @@ class ControlLoop:
         hpa_spec: Optional[HpaSpec] = None,
         hpa_metric: str = REFERENCE_RULE_NAME,
         extra_samples: Optional[Callable[[], List[Sample]]] = None,
+        use_adapter: bool = False,
+        adapter_target: tuple = ("default", "deployments", "cuda-test"),
     ):
         self.scraper = scraper
         self.rules = rules if rules is not None else [
@@ -73,6 +77,13 @@ class ControlLoop:
         self.hpa_state = HpaState()
         self.hpa_metric = hpa_metric
         self.extra_samples = extra_samples
+        # use_adapter=True routes the HPA's metric fetch through the
+        # prometheus-adapter default-rule model (adapter.py) — the full
+        # L4 hop, discovery included — instead of reading the recorded
+        # value directly. adapter_target is the HPA scaleTargetRef
+        # (namespace, resource, name), reference cuda-test-hpa.yaml:13-20.
+        self.adapter = Adapter() if use_adapter else None
+        self.adapter_target = adapter_target
         self.recorded_series: List[Sample] = []
         # rule name -> first error message (a failing rule is reported once,
         # like Prometheus's unhealthy-rule state)
@@ -109,7 +120,17 @@ class ControlLoop:
             recorded[rule.record] = vec[0].value if len(vec) == 1 else None
         t2 = time.monotonic()
 
-        metric_value = recorded.get(self.hpa_metric)
+        if self.adapter is not None:
+            # L4 for real: adapter discovery + Object-metric GET over the
+            # recorded series, exactly what the HPA controller reads
+            # 
+            ns, resource, name = self.adapter_target
+            self.adapter.update(self.recorded_series)
+            metric_value = self.adapter.get_object_metric_value(
+                ns, resource, name, self.hpa_metric)
+        else:
+            metric_value = recorded.get(self.hpa_metric)
+        t2b = time.monotonic()
         replicas = reconcile(
             self.hpa_spec, self.hpa_state, metric_value,
             now_s if now_s is not None else time.time(),
@@ -119,11 +140,12 @@ class ControlLoop:
         return LoopResult(
             scrape_s=t1 - t0,
             rule_eval_s=t2 - t1,
-            hpa_s=t3 - t2,
+            hpa_s=t3 - t2b,
             total_s=t3 - t0,
             recorded=recorded,
             replicas=replicas,
             metric_value=metric_value,
+            adapter_s=t2b - t2,
         )
 
 
