@@ -5,10 +5,23 @@ pipeline on MI355X, under real MFMA bf16 GEMM load.
 Measures BASELINE.json's headline metric: **p50 scrape->HPA-scale latency**
 — the time for one full control cycle (scrape the native exporter's
 /metrics for every GPU -> evaluate the reference recording rule
-(cuda-test-prometheusrule.yaml:13 semantics) -> HPA reconcile decision) —
-while every GPU runs the CDNA4 MFMA bf16 GEMM load generator at a high duty
-cycle. Also reports the GPU-util metric error vs rocm-smi (the second
-north-star number) in `config.util_err_pct`.
+(cuda-test-prometheusrule.yaml:13 semantics) -> prometheus-adapter
+default-rule Object-metric GET -> HPA reconcile decision) — while every
+GPU runs the CDNA4 MFMA bf16 GEMM load generator at a high duty cycle.
+Also reports the GPU-util metric error vs rocm-smi (the second north-star
+number) in `config.util_err_vs_rocm_smi_pct`.
+
+Freshness-honest (round-1 verdict): every timed step first waits for the
+exporter's sample counter (`amd_exporter_samples_total`) to ADVANCE, so
+each of the K cycles decides on a fresh sample — no step re-reads stale
+data. `value` is the p50 of the per-cycle work time (scrape+rule+adapter+
+HPA); `ms_per_step` is wall/K and therefore ~ the exporter interval by
+construction. Self-evidencing: `config.observed_busy_pct` is the mean GPU
+utilization the control loop itself saw across the timed steps, and the
+run FAILS on a GPU box if that is below half the load target — the record
+proves its own load. `config.load_step_detection_s` is the end-to-end
+falling-edge latency: stop the load, measure time until the autoscale
+metric drops below half target (the soak's detection metric, in-run).
 
 Reference baseline: the cadence/latency parameters in BASELINE.md — 10 s
 exporter tick, <=30 s to metric availability (no published latency number,
@@ -152,7 +165,8 @@ def main():
             "w", suffix=".csv", delete=False, prefix="bench-metrics-"
         )
         mf.write("dcgm_gpu_utilization\ndcgm_gpu_temp\n"
-                 "amd_xgmi_total_bytes_per_second\n")
+                 "amd_xgmi_total_bytes_per_second\n"
+                 "amd_exporter_samples_total\n")
         mf.close()
         kw = dict(interval_ms=args.exporter_interval_ms, metric_file=mf.name)
         if has_gpu:
@@ -181,6 +195,7 @@ def main():
             scraper,
             hpa_spec=HpaSpec(min_replicas=1, max_replicas=8, target_value=5.0),
             extra_samples=lambda: synth_pod_labels(pods),
+            use_adapter=True,  # L4 for real: default-rule discovery + GET
         )
 
     def barrier():
@@ -190,6 +205,36 @@ def main():
     def sync():
         if has_gpu:
             torch.cuda.synchronize()
+
+    # --- freshness: wait until the exporter's per-device sample counter
+    # advances past `last`, so the next control cycle reads a NEW sample
+    # (round-1 verdict: successive steps must not re-read the same tick).
+    import urllib.request
+
+    from mi355x_gpu_hpa.control import parse_prometheus_text
+
+    def exporter_tick(timeout_s=2.0):
+        try:
+            with urllib.request.urlopen(exporter.url, timeout=timeout_s) as r:
+                text = r.read().decode()
+        except Exception:  # noqa: BLE001
+            return None
+        ticks = [s.value for s in parse_prometheus_text(text)
+                 if s.name == "amd_exporter_samples_total"]
+        return max(ticks) if ticks else None
+
+    stale_steps = 0
+
+    def wait_fresh_tick(last):
+        nonlocal stale_steps
+        deadline = time.monotonic() + 5 * args.exporter_interval_ms / 1e3
+        while time.monotonic() < deadline:
+            t = exporter_tick()
+            if t is not None and (last is None or t > last):
+                return t
+            time.sleep(args.exporter_interval_ms / 1e3 / 10)
+        stale_steps += 1    # exporter stalled: proceed, but say so
+        return last
 
     # --- warmup -----------------------------------------------------------
     # let the exporter take >=2 samples so windowed rates exist
@@ -203,21 +248,28 @@ def main():
     barrier()
     sync()
 
-    # --- timed region: exactly K control-loop steps, with one xGMI
-    # all-reduce per step when world > 1 -----------------------------------
+    # --- timed region: exactly K tick-aligned control-loop steps, with one
+    # xGMI all-reduce per step when world > 1 ------------------------------
     latencies = []
+    busy_seen = []          # the metric value each cycle decided on
+    last_tick = exporter_tick() if rank == 0 else None
     t0 = time.monotonic()
     for _ in range(args.steps):
         w = xgmi_tick()
         if rank == 0:
+            last_tick = wait_fresh_tick(last_tick)
             r = loop.step()
             latencies.append(r.total_s)
+            if r.metric_value is not None:
+                busy_seen.append(r.metric_value)
         if w:
             w.wait()
     barrier()
     sync()
     t1 = time.monotonic()
     wall_s = t1 - t0
+    stale_timed = stale_steps   # snapshot before the detection phase reuses
+                                # wait_fresh_tick
 
     # max over ranks (non-zero ranks have ~0 step time; the max is rank 0's)
     if world > 1:
@@ -230,14 +282,11 @@ def main():
     # single snapshot pair jitters under a duty-cycled load; take the median
     # over several paired reads.
     if rank == 0 and has_gpu:
-        parse = __import__(
-            "mi355x_gpu_hpa.control", fromlist=["parse_prometheus_text"]
-        ).parse_prometheus_text
         per_gpu_errs = {}
         samples = []
         for _ in range(5):
             oracle = rocm_smi_busy()
-            samples = parse(exporter.scrape())
+            samples = parse_prometheus_text(exporter.scrape())
             ours = {
                 int(s.labels["gpu"]): s.value
                 for s in samples
@@ -259,6 +308,29 @@ def main():
         if xgmi_bps and max(xgmi_bps) > 0:
             log(f"xGMI total traffic: {max(xgmi_bps)/1e9:.2f} GB/s (max GPU)")
 
+    # --- load-step detection: stop the load and measure the end-to-end
+    # falling-edge latency (exporter tick -> scrape -> rule -> adapter ->
+    # decision sees the drop). This folds the soak's detection metric
+    # (profiles/control_loop_soak.md) into the bench record itself.
+    load_step_detection_s = None
+    detect_threshold = args.load_util / 2.0
+    if rank == 0 and has_gpu and load_thread:
+        t_stop = time.monotonic()
+        stop_flag.value = 1
+        deadline = t_stop + 15.0
+        lt = exporter_tick()
+        while time.monotonic() < deadline:
+            lt = wait_fresh_tick(lt)
+            r = loop.step()
+            if r.metric_value is not None and r.metric_value < detect_threshold:
+                load_step_detection_s = time.monotonic() - t_stop
+                break
+        if load_step_detection_s is not None:
+            log(f"load-step (fall to <{detect_threshold:.0f}%) detected in "
+                f"{load_step_detection_s*1e3:.0f} ms end-to-end")
+        else:
+            log("load-step NOT detected within 15 s")
+
     # --- stop load + report ----------------------------------------------
     stop_flag.value = 1
     if load_thread:
@@ -272,6 +344,7 @@ def main():
             pass
 
     if rank == 0:
+        observed_busy = statistics.mean(busy_seen) if busy_seen else None
         latencies.sort()
         p50_ms = statistics.median(latencies) * 1e3
         p99_ms = latencies[min(len(latencies) - 1, int(len(latencies) * 0.99))] * 1e3
@@ -301,9 +374,32 @@ def main():
                 "gpus_exported": n_exported,
                 "final_replicas": final_replicas,
                 "reference_cadence_s": 10.0,
+                # self-evidencing (round-1 verdict): the load as the timed
+                # control cycles themselves observed it, tick alignment
+                # stats, and the in-run end-to-end detection latency
+                "observed_busy_pct": (round(observed_busy, 2)
+                                      if observed_busy is not None else None),
+                "tick_aligned_steps": args.steps - stale_timed,
+                "stale_steps": stale_timed,
+                "load_step_detection_s": (round(load_step_detection_s, 4)
+                                          if load_step_detection_s is not None
+                                          else None),
             },
         }
         print(json.dumps(result), flush=True)
+
+        # the record must prove its own load: on a GPU box, a bench that
+        # ran with the GPUs idle (or an exporter that never saw the burn)
+        # is invalid — fail loudly rather than emit a hollow number
+        if has_gpu and (observed_busy is None
+                        or observed_busy < args.load_util / 2.0):
+            log(f"FAIL: observed busy {observed_busy} < half the "
+                f"{args.load_util}% load target — load not evidenced")
+            sys.exit(1)
+        if has_gpu and stale_timed > args.steps // 4:
+            log(f"FAIL: {stale_timed}/{args.steps} steps ran on stale "
+                "exporter data")
+            sys.exit(1)
 
     if world > 1:
         dist.destroy_process_group()

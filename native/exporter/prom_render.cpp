@@ -224,6 +224,12 @@ std::string render_metrics(const std::vector<DeviceMetrics>& devs,
     w.family("amd_gpu_hotspot_temp", "Junction/hotspot temperature (C).", "gauge");
     for (auto& d : devs) w.sample(d, attr, d.sample.temp_hotspot_c);
 
+    w.family("amd_exporter_samples_total",
+             "Sampling ticks taken for this device since exporter start "
+             "(freshness counter: advances once per -c interval).",
+             "counter");
+    for (auto& d : devs) w.sample(d, attr, (double)d.samples_taken);
+
     return w.str();
 }
 

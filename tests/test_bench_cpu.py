@@ -43,6 +43,17 @@ def test_bench_single():
     check_contract(r, 1)
     # with the mock exporter the whole loop must be comfortably sub-10ms
     assert r["value"] < 100
+    # self-evidencing fields (round-1 verdict item 3)
+    cfg = r["config"]
+    assert "observed_busy_pct" in cfg
+    assert "load_step_detection_s" in cfg
+    # tick alignment: every timed step must have seen a fresh exporter
+    # sample (the mock exporter ticks reliably on CPU)
+    assert cfg["tick_aligned_steps"] == 10
+    assert cfg["stale_steps"] == 0
+    # ms_per_step is now tick-bound: ~the exporter interval (100 ms), far
+    # above the per-cycle latency — the freshness-honest shape
+    assert r["ms_per_step"] >= 50
 
 
 def test_bench_world2_gloo():
@@ -55,3 +66,19 @@ def test_bench_world2_gloo():
     assert p.returncode == 0, p.stderr.decode()[-2000:]
     r = _last_json_line(p.stdout)
     check_contract(r, 2)
+
+
+def test_bench_world8_gloo():
+    """The exact N=8 launch shape the driver uses for SCALE (round-1
+    verdict item 9: keep multi-GPU readiness warm — world-8 rendezvous,
+    8-device mock exporter, per-rank loop roles must not bit-rot)."""
+    p = _run([
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+        "--master-port", "29519", "bench.py",
+        "--gpus", "8", "--steps", "3", "--warmup", "1",
+    ], timeout=420)
+    assert p.returncode == 0, p.stderr.decode()[-2000:]
+    r = _last_json_line(p.stdout)
+    check_contract(r, 8)
+    assert r["config"]["gpus_exported"] in (0, 8)  # 8 once on-GPU validated
