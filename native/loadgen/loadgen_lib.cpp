@@ -388,13 +388,19 @@ static double read_busy_pct(const std::string& path)
 
 // Duty-cycled GEMM burn: aim at `target_util_pct` GPU-busy for `seconds`.
 // Duty cycle over a `period_ms` window: run GEMM launches for duty*period,
-// sleep the rest. The duty fraction is CLOSED-LOOP: each period reads the
-// measured amdgpu busy% (gpu_busy_percent sysfs, the counter the exporter
-// and rocm-smi report) and trims the duty with slow integral action, so
-// launch gaps and the firmware's busy-accounting bias are corrected
-// instead of left to drift (round-1 verdict: the open-loop burn needed a
-// +/-25pp test band; closed-loop targets +/-10pp). Without sysfs (no
-// /sys in a sandbox) it degrades to the open-loop behavior.
+// sleep the rest. Two mechanisms close the gap between wall-clock duty and
+// the GRBM busy% that rocm-smi/the exporter report (round 1 measured an
+// open-loop undershoot of ~0.72x: launch+sync gaps inside a "busy" burst
+// are GPU-idle wall time):
+//   * launches are BATCHED (4 queued back-to-back per sync) so intra-burst
+//     gaps mostly vanish;
+//   * the duty fraction is CLOSED-LOOP on the measured GPU-active time of
+//     each burst (hipEvent elapsed time — the same kernel-resident time
+//     GRBM counts), trimmed with bounded integral action. As a secondary
+//     oracle the amdgpu gpu_busy_percent sysfs is blended in when
+//     readable, which also accounts for foreign load on the device.
+// Round-1 verdict item 8: the open-loop burn needed a +/-25pp test band;
+// closed-loop targets +/-10pp.
 // stop_flag: optional; polled between periods (set non-zero to stop early).
 int lg_gemm_burn(int device, double target_util_pct, double seconds,
                  int m, int n, int k, double period_ms, volatile int* stop_flag)
@@ -409,6 +415,9 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
+    hipEvent_t ev0, ev1;
+    LG_CHECK(hipEventCreate(&ev0));
+    LG_CHECK(hipEventCreate(&ev1));
     // one calibration launch so the first period isn't all compile/warmup
     gemm_launch(g, 0, burn_variant);
     LG_CHECK(hipDeviceSynchronize());
@@ -420,31 +429,43 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
     // must not drag it to zero)
     const double duty_lo = duty > 0.25 ? duty - 0.25 : 0.0;
     const double duty_hi = duty + 0.25 < 1.0 ? duty + 0.25 : 1.0;
-    const double kI = 0.0008;  // duty fraction per %-error per period
-    double ema = -1;           // smoothed measured busy% (counter jitters)
+    const double kI = 0.004;   // duty fraction per %-error per period
+    double ema = -1;           // smoothed measured active% over the period
 
     double t_end = now_ms() + seconds * 1e3;
     while (now_ms() < t_end) {
         if (stop_flag && *stop_flag) break;
         double period_start = now_ms();
         double busy_until = period_start + period_ms * duty;
+        float active_ms = 0;
         while (now_ms() < busy_until) {
-            gemm_launch(g, 0, burn_variant);
+            LG_CHECK(hipEventRecord(ev0, 0));
+            for (int b = 0; b < 4; ++b) gemm_launch(g, 0, burn_variant);
+            LG_CHECK(hipEventRecord(ev1, 0));
             LG_CHECK(hipDeviceSynchronize());
+            float dt = 0;
+            LG_CHECK(hipEventElapsedTime(&dt, ev0, ev1));
+            active_ms += dt;
         }
         if (target_util_pct > 0 && target_util_pct < 100) {
-            double meas = read_busy_pct(busy_path);
-            if (meas >= 0) {
-                ema = ema < 0 ? meas : 0.8 * ema + 0.2 * meas;
-                duty += kI * (target_util_pct - ema);
-                if (duty < duty_lo) duty = duty_lo;
-                if (duty > duty_hi) duty = duty_hi;
-            }
+            double active_pct = active_ms / period_ms * 100.0;
+            double sysfs_pct = read_busy_pct(busy_path);
+            // prefer the device-global sysfs busy% when available (it sees
+            // foreign load too); the event measurement is the fallback
+            double meas = sysfs_pct >= 0
+                              ? 0.5 * sysfs_pct + 0.5 * active_pct
+                              : active_pct;
+            ema = ema < 0 ? meas : 0.7 * ema + 0.3 * meas;
+            duty += kI * (target_util_pct - ema);
+            if (duty < duty_lo) duty = duty_lo;
+            if (duty > duty_hi) duty = duty_hi;
         }
         double rest = period_start + period_ms - now_ms();
         if (rest > 0)
             std::this_thread::sleep_for(std::chrono::duration<double, std::milli>(rest));
     }
+    (void)hipEventDestroy(ev0);
+    (void)hipEventDestroy(ev1);
     gemm_free(g);
     return 0;
 }
