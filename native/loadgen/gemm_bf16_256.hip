@@ -748,3 +748,208 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_soft(
 {
     gemm_bf16_tn_256_impl<6, 1, 0>(A, Bt, C, M, N, K, tiles_per_cta);
 }
+
+// ---------------------------------------------------------------------------
+// d9: ONE barrier per K-tile — the whole tile is a single scheduling window.
+//
+// The round-1 PMC bound: 34% of wave time parked at the 8 per-K-tile
+// barriers/waits, MFMA pipe ~50% busy (profiles/gemm_bf16_256_ladder.md),
+// and the d7 experiment showed shaving barriers while keeping per-phase
+// read drains is not the lever. The d9 observation is a liveness fact: if
+// EVERY half (A0,A1,B0,B1) of K-tile kt+1 stages into buffer buf^1, then
+// every staged slot has been dead since the kt-1 -> kt boundary barrier
+// (buf^1 was fully consumed during tile kt-1; B is read only at q0, A by
+// q3, and all reads retire before their consuming MFMAs which precede
+// that barrier). So NO mid-tile publish points are needed at all:
+//
+//     boundary barrier
+//       q0: read B(all 4)+A(2) | stage kt+1 A0,A1 -> buf^1 | MFMA q0
+//       q1: read A(2)          | stage kt+1 B0    -> buf^1 | MFMA q1
+//       q2: read A(2)          | stage kt+1 B1    -> buf^1 | MFMA q2
+//       q3: read A(2)          |                           | MFMA q3
+//     s_waitcnt vmcnt(0)   (kt+1 landed; DMAs had ~3 phases >= 1.5 us)
+//     boundary barrier
+//
+// with NO fences between the phases: the compiler schedules 64 MFMAs,
+// 24 ds_reads and 6 glds of a tile as one block, so each phase's LDS
+// reads issue under the previous phase's MFMA segment (counted lgkm
+// waits), and co-resident waves drift freely within the tile — the
+// partner-overlap the barrier-locked d6 could never reach. The staging
+// depth shrinks from d6's two-tiles-ahead to one, which is why the
+// drain can be a full vmcnt(0): by q3 the q0-issued DMAs are ~1.5 us
+// old vs ~0.4 us HBM latency.
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl9(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;
+    const int wc = w & 3;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    auto frag_off = [&](int row_in_half, int ks) {
+        return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
+    };
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[2][2];
+        bf16x8 bfrag[4][2];
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned short* la = &lds[(buf * 4 + wr) * HALF_HW];
+            const unsigned short* lb = &lds[(buf * 4 + 2 + (wc >> 1)) * HALF_HW];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = q * 2;
+#pragma unroll
+                for (int m = 0; m < 2; ++m) {
+                    const int row = (mbase + m) * 16 + (lane & 15);
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        afrag[m][ks] =
+                            *(const bf16x8*)((const char*)la + frag_off(row, ks));
+                }
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 4; ++n) {
+                        const int col = bcol0 + n * 16 + (lane & 15);
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                            frag_off(col, ks));
+                    }
+                }
+
+                // next tile's halves -> buf^1 (all slots dead since the
+                // previous boundary barrier; no publish needed mid-tile)
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 1, 2, buf ^ 1);
+                } else if (q == 2) {
+                    stage(kt + 1, 3, buf ^ 1);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            acc[mbase + m][n] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag[m][ks], bfrag[n][ks],
+                                    acc[mbase + m][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            // single boundary: land kt+1's four halves, then publish
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl9<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// d9 without the super-tile raster (isolates the schedule effect).
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9nr(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl9<0>(A, Bt, C, M, N, K, tiles_per_cta);
+}

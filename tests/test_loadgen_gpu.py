@@ -86,6 +86,30 @@ def test_gemm_bf16_256_numerics(gpu, m, n, k):
     np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 256, 64), (256, 256, 128),
+                                   (512, 256, 256), (512, 512, 1024),
+                                   (1024, 1024, 2048)])
+def test_gemm_bf16_256_d9_numerics(gpu, m, n, k):
+    """The single-barrier-per-K-tile d9 schedule vs torch fp32 (multi-K-tile
+    shapes exercise the all-four-halves-ahead staging; k=64 the clamped
+    tail). Same accumulation order as the product kernel, so also bitwise-
+    comparable to it."""
+    import torch
+
+    lg = _loadgen()
+    rng = np.random.default_rng(11)
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
+    a[:, 0] += np.arange(m) * 0.01
+    bt[:, 0] -= np.arange(n) * 0.01
+    got = lg.gemm_bf16(a, bt, variant=11)
+    ref = (torch.from_numpy(a).bfloat16().float()
+           @ torch.from_numpy(bt).bfloat16().float().T).numpy()
+    np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
+    same = lg.gemm_bf16(a, bt, variant=2)
+    np.testing.assert_allclose(got, same, rtol=1e-6, atol=1e-5)
+
+
 def test_gemm_bf16_256_matches_128(gpu):
     """Cross-check: both kernels compute identical bf16 sums (same
     accumulation order over K) — results should agree to fp32 rounding."""
