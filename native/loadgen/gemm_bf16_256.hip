@@ -953,3 +953,174 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9nr(
 {
     gemm_bf16_tn_256_impl9<0>(A, Bt, C, M, N, K, tiles_per_cta);
 }
+
+// ---------------------------------------------------------------------------
+// d14: 16-wave CTA (1024 threads), 64x64 wave tiles, 4 waves/SIMD.
+//
+// The occupancy lever d9 cannot reach: with 128 acc VGPRs per wave the
+// 128x64-tile schedules cap at 2 waves/SIMD (512-VGPR file), and the
+// measured MFMA-pipe busy ~40-50% is then bounded by per-wave stall
+// fraction. Shrinking the wave tile to 64x64 (4x4 wave grid over the same
+// 256x256 CTA tile) drops the budget to acc 64 + B-frags 32 + A-frags 8
+// ~= 104+addressing VGPRs — under the 128 needed for FOUR waves per SIMD,
+// so the SIMD holds 2x the MFMA issue sources at every point. Same LDS
+// image/swizzle/glds machinery (16 waves -> 1 KiB piece per wave), same
+// single-barrier-per-K-tile liveness as d9. LDS read traffic doubles per
+// FLOP vs d9 (B re-read by 4 waves per column block instead of 2) to
+// ~78% of LDS bandwidth at 2 PF/s — the conflict-free swizzle is what
+// makes that budget real.
+// __launch_bounds__(1024, 4): the 4-waves/SIMD guarantee (<=128 VGPRs,
+// enforced at compile time — check the .sdata VGPR count, do not assume).
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl14(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;       // 0..15
+    const int lane = tid & 63;
+    const int wr = w >> 2;        // 0..3: A rows wr*64..+64
+    const int wc = w & 3;         // 0..3: B cols wc*64..+64
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    auto frag_off = [&](int row_in_half, int ks) {
+        return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
+    };
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[4][4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        // one 1-KiB glds piece per wave (16 waves x 1 KiB = one half-tile)
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)
+                    (src + (long)(w * 8 + src_row) * K),
+                (__attribute__((address_space(3))) unsigned int*)(dst + w * 512),
+                16, 0, 0);
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[2];
+        bf16x8 bfrag[4][2];
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned short* la = &lds[(buf * 4 + (wr >> 1)) * HALF_HW];
+            const unsigned short* lb = &lds[(buf * 4 + 2 + (wc >> 1)) * HALF_HW];
+            const int arow0 = (wr & 1) * 64;
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int row = arow0 + q * 16 + (lane & 15);
+#pragma unroll
+                for (int ks = 0; ks < 2; ++ks)
+                    afrag[ks] =
+                        *(const bf16x8*)((const char*)la + frag_off(row, ks));
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 4; ++n) {
+                        const int col = bcol0 + n * 16 + (lane & 15);
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                            frag_off(col, ks));
+                    }
+                }
+
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 1, 2, buf ^ 1);
+                } else if (q == 2) {
+                    stage(kt + 1, 3, buf ^ 1);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int n = 0; n < 4; ++n)
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        acc[q][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            afrag[ks], bfrag[n][ks], acc[q][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 64 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(1024, 4) gemm_bf16_tn_256_d14(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl14<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}

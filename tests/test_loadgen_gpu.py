@@ -110,6 +110,27 @@ def test_gemm_bf16_256_d9_numerics(gpu, m, n, k):
     np.testing.assert_allclose(got, same, rtol=1e-6, atol=1e-5)
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 256, 64), (256, 256, 128),
+                                   (512, 512, 1024), (1024, 1024, 2048)])
+def test_gemm_bf16_256_d14_numerics(gpu, m, n, k):
+    """The 16-wave 4-waves/SIMD kernel vs torch fp32 + bitwise vs the
+    product kernel (same accumulation order)."""
+    import torch
+
+    lg = _loadgen()
+    rng = np.random.default_rng(14)
+    a = rng.uniform(-1, 1, (m, k)).astype(np.float32)
+    bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
+    a[:, 0] += np.arange(m) * 0.01
+    bt[:, 0] -= np.arange(n) * 0.01
+    got = lg.gemm_bf16(a, bt, variant=13)
+    ref = (torch.from_numpy(a).bfloat16().float()
+           @ torch.from_numpy(bt).bfloat16().float().T).numpy()
+    np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
+    same = lg.gemm_bf16(a, bt, variant=2)
+    np.testing.assert_allclose(got, same, rtol=1e-6, atol=1e-5)
+
+
 def test_gemm_bf16_256_matches_128(gpu):
     """Cross-check: both kernels compute identical bf16 sums (same
     accumulation order over K) — results should agree to fp32 rounding."""

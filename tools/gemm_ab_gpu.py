@@ -5,7 +5,8 @@ Usage: python tools/gemm_ab_gpu.py [variants...] [--shapes 8192 16k8k]
 Runs each variant's bench entry several times interleaved (A,B,A,B,...) so
 DVFS drift hits all variants equally; prints per-run TF/s and the max.
 Variant map: 2=product d6+raster, 7=d7 5-barrier, 8=d6 no-raster,
-10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster.
+10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster, 13=d14
+16-wave 4-waves/SIMD.
 """
 
 import json
