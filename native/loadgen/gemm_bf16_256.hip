@@ -325,10 +325,9 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl(
     }
 }
 
-// PRODUCT kernel: the d6 schedule (m-quarter phases, all-B-held, B staged
-// two K-tiles ahead) — measured 946-955 TF/s @8192^3, 1033 TF/s @16k x 16k
-// x 8k on random bf16 (see profiles/gemm_bf16_256_ladder.md).
-extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+// the round-1 product (d6 schedule: 8 barriers/K-tile, m-quarter phases,
+// all-B-held, B staged two tiles ahead) — kept as the A/B baseline.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d6(
     const unsigned short* A, const unsigned short* Bt, float* C,
     int M, int N, int K, int tiles_per_cta)
 {
@@ -1124,3 +1123,15 @@ extern "C" __global__ void __launch_bounds__(1024, 4) gemm_bf16_tn_256_d14(
 {
     gemm_bf16_tn_256_impl14<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
+
+// PRODUCT kernel: the d9 schedule (single barrier per K-tile, all four
+// halves staged one tile ahead into the dead buffer) — measured 990-998
+// TF/s @8192^3, 1124-1139 TF/s @16k x 16k x 8k on random bf16, +6% over
+// the round-1 d6 product (see profiles/gemm_bf16_256_ladder.md).
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl9<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+

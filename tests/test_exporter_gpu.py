@@ -77,7 +77,10 @@ def test_util_error_vs_rocm_smi_under_load(gpu):
     assert 0 in ours and 0 in oracle, (ours, oracle)
     # device 0 is under continuous GEMM load: both must read high
     assert ours[0] > 80, f"exporter busy {ours[0]}% under full load"
-    assert abs(ours[0] - oracle[0]) <= 15, (
+    # tightened from 15pp (round-1 verdict item 8): under a CONTINUOUS
+    # (100% duty) load there is no window jitter; bench runs measured
+    # 2-5pp paired error (BENCH_r01/r02)
+    assert abs(ours[0] - oracle[0]) <= 8, (
         f"exporter {ours[0]}% vs rocm-smi {oracle[0]}%"
     )
 
@@ -110,8 +113,11 @@ def test_duty_cycle_tracks_target(gpu):
         stop.value = 1
         t.join(timeout=15)
     mean = sum(vals) / len(vals)
-    # generous band: rsmi busy sampling vs our open-loop duty cycle
-    assert 25 <= mean <= 75, f"mean busy {mean}% for 50% duty target ({vals})"
+    # +/-10pp band (round-1 verdict item 8): the closed-loop duty
+    # controller (event-measured GPU-active feedback) landed 48.6/78.0/22.4
+    # for targets 50/80/20 across boxes (profiles/duty_closed_loop.md);
+    # the band covers instantaneous-sample jitter (stdev ~7pp, n=8+).
+    assert 40 <= mean <= 60, f"mean busy {mean}% for 50% duty target ({vals})"
 
 
 def test_live_schema_superset_of_fixtures(gpu):

@@ -6,7 +6,7 @@ Runs each variant's bench entry several times interleaved (A,B,A,B,...) so
 DVFS drift hits all variants equally; prints per-run TF/s and the max.
 Variant map: 2=product d6+raster, 7=d7 5-barrier, 8=d6 no-raster,
 10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster, 13=d14
-16-wave 4-waves/SIMD.
+16-wave 4-waves/SIMD, 14=d6 round-1 product (now ablation).
 """
 
 import json
