@@ -778,7 +778,13 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_soft(
 // drain can be a full vmcnt(0): by q3 the q0-issued DMAs are ~1.5 us
 // old vs ~0.4 us HBM latency.
 // ---------------------------------------------------------------------------
-template <int RASTER = 1>
+// WIDE_EPI: per-wave LDS transpose of the accumulators so C stores are
+// global_store_dwordx4 (4 rows x 64 B -> 16 rows x 64 B per instruction,
+// 4x fewer store instructions). The 128 KiB tile LDS is dead after the
+// K-loop's final barrier; each wave uses a private padded 16x68-f32
+// region (bank-conflict-free for both the scatter writes and the b128
+// row reads; pad 68 makes bank = 4*row + col mod 64 a permutation).
+template <int RASTER = 1, int WIDE_EPI = 0>
 __device__ __forceinline__ void gemm_bf16_tn_256_impl9(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
@@ -922,15 +928,41 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl9(
             __builtin_amdgcn_s_barrier();
         }
 
+        if (WIDE_EPI) {
+            float* scr = (float*)lds + w * 1088; // private 16 x 68 f32
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
+            for (int i = 0; i < 8; ++i) {
+                if (i)  // WAR: chunk i-1's reads must retire first
+                    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
-            for (int j = 0; j < 4; ++j) {
+                for (int j = 0; j < 4; ++j)
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
-                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
-                    C[row * (long)N + col] = acc[i][j][r];
+                    for (int r = 0; r < 4; ++r)
+                        scr[((lane >> 4) * 4 + r) * 68 + j * 16 + (lane & 15)] =
+                            acc[i][j][r];
+                // RAW: DS ops are per-wave in-order; drain before reads
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                const long row = row0 + wr * 128 + i * 16 + (lane & 15);
+                float* dst = C + row * (long)N + col0 + wc * 64;
+#pragma unroll
+                for (int rq = 0; rq < 4; ++rq) {
+                    const int cq = (lane >> 4) + rq * 4;
+                    f32x4 v = *(const f32x4*)&scr[(lane & 15) * 68 + cq * 4];
+                    *(f32x4*)(dst + cq * 4) = v;
+                }
+            }
+        } else {
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        const long row =
+                            row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                        const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                        C[row * (long)N + col] = acc[i][j][r];
+                    }
                 }
             }
         }
@@ -943,6 +975,14 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl9<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// d9 + wide epilogue (LDS-transposed dwordx4 stores) — A/B candidate.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9w(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl9<1, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
 // d9 without the super-tile raster (isolates the schedule effect).

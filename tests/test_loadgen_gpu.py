@@ -86,10 +86,11 @@ def test_gemm_bf16_256_numerics(gpu, m, n, k):
     np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
 
 
+@pytest.mark.parametrize("variant", [11, 15])  # d9, d9+wide-epilogue
 @pytest.mark.parametrize("m,n,k", [(256, 256, 64), (256, 256, 128),
                                    (512, 256, 256), (512, 512, 1024),
                                    (1024, 1024, 2048)])
-def test_gemm_bf16_256_d9_numerics(gpu, m, n, k):
+def test_gemm_bf16_256_d9_numerics(gpu, m, n, k, variant):
     """The single-barrier-per-K-tile d9 schedule vs torch fp32 (multi-K-tile
     shapes exercise the all-four-halves-ahead staging; k=64 the clamped
     tail). Same accumulation order as the product kernel, so also bitwise-
@@ -102,7 +103,7 @@ def test_gemm_bf16_256_d9_numerics(gpu, m, n, k):
     bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
     a[:, 0] += np.arange(m) * 0.01
     bt[:, 0] -= np.arange(n) * 0.01
-    got = lg.gemm_bf16(a, bt, variant=11)
+    got = lg.gemm_bf16(a, bt, variant=variant)
     ref = (torch.from_numpy(a).bfloat16().float()
            @ torch.from_numpy(bt).bfloat16().float().T).numpy()
     np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
