@@ -784,7 +784,9 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_soft(
 // K-loop's final barrier; each wave uses a private padded 16x68-f32
 // region (bank-conflict-free for both the scatter writes and the b128
 // row reads; pad 68 makes bank = 4*row + col mod 64 a permutation).
-template <int RASTER = 1, int WIDE_EPI = 0>
+// STAGE_Q0: issue all four kt+1 stages at q0 (instead of spreading over
+// q0-q2) — maximizes DMA landing margin before the boundary vmcnt(0).
+template <int RASTER = 1, int WIDE_EPI = 0, int STAGE_Q0 = 0>
 __device__ __forceinline__ void gemm_bf16_tn_256_impl9(
     const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
@@ -904,9 +906,13 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl9(
                 if (q == 0) {
                     stage(kt + 1, 0, buf ^ 1);
                     stage(kt + 1, 1, buf ^ 1);
-                } else if (q == 1) {
+                    if (STAGE_Q0) {
+                        stage(kt + 1, 2, buf ^ 1);
+                        stage(kt + 1, 3, buf ^ 1);
+                    }
+                } else if (!STAGE_Q0 && q == 1) {
                     stage(kt + 1, 2, buf ^ 1);
-                } else if (q == 2) {
+                } else if (!STAGE_Q0 && q == 2) {
                     stage(kt + 1, 3, buf ^ 1);
                 }
 
@@ -983,6 +989,14 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9w(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_bf16_tn_256_impl9<1, 1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// d9 with all four stages issued at q0 — A/B candidate.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9e(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl9<1, 0, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
 // d9 without the super-tile raster (isolates the schedule effect).
@@ -1122,9 +1136,13 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl14(
                 if (q == 0) {
                     stage(kt + 1, 0, buf ^ 1);
                     stage(kt + 1, 1, buf ^ 1);
-                } else if (q == 1) {
+                    if (STAGE_Q0) {
+                        stage(kt + 1, 2, buf ^ 1);
+                        stage(kt + 1, 3, buf ^ 1);
+                    }
+                } else if (!STAGE_Q0 && q == 1) {
                     stage(kt + 1, 2, buf ^ 1);
-                } else if (q == 2) {
+                } else if (!STAGE_Q0 && q == 2) {
                     stage(kt + 1, 3, buf ^ 1);
                 }
 
