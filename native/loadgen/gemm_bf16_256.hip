@@ -1136,13 +1136,9 @@ __device__ __forceinline__ void gemm_bf16_tn_256_impl14(
                 if (q == 0) {
                     stage(kt + 1, 0, buf ^ 1);
                     stage(kt + 1, 1, buf ^ 1);
-                    if (STAGE_Q0) {
-                        stage(kt + 1, 2, buf ^ 1);
-                        stage(kt + 1, 3, buf ^ 1);
-                    }
-                } else if (!STAGE_Q0 && q == 1) {
+                } else if (q == 1) {
                     stage(kt + 1, 2, buf ^ 1);
-                } else if (!STAGE_Q0 && q == 2) {
+                } else if (q == 2) {
                     stage(kt + 1, 3, buf ^ 1);
                 }
 
