@@ -43,6 +43,7 @@ class ExporterProcess:
         metric_file: Optional[str] = None,
         mock_busy_file: Optional[str] = None,
         gpu_id_type: Optional[str] = None,
+        backend: Optional[str] = None,
     ):
         self.args = [EXPORTER_BIN, "-c", str(interval_ms)]
         # port 0 would race; pick a free one ourselves
@@ -67,6 +68,8 @@ class ExporterProcess:
             self.args += ["--mock-busy-file", mock_busy_file]
         if gpu_id_type:
             self.args += ["--kubernetes-gpu-id-type", gpu_id_type]
+        if backend:
+            self.args += ["--backend", backend]
         self.proc: Optional[subprocess.Popen] = None
 
     @property

@@ -17,6 +17,8 @@ namespace mi355x {
 
 constexpr int kMaxXgmiLinks = 8;
 
+constexpr int kMaxXcp = 8;   // compute partitions per GPU (SPX..CPX)
+
 struct GpuSample {
     bool ok = false;              // sample succeeded for this device
     double ts_ms = 0;             // host steady-clock timestamp
@@ -73,6 +75,13 @@ struct GpuSample {
 
     // PCIe replay events (accumulated)
     double pcie_replay_count = -1;
+
+    // per-XCP (compute partition) instantaneous gfx busy %, mean over the
+    // partition's XCCs (gpu_metrics v1.6+ xcp_stats). num_partitions = 0
+    // when the firmware reports none (then whole-GPU busy_pct is the only
+    // utilization series).
+    int num_partitions = 0;
+    double xcp_busy_pct[kMaxXcp] = {-1, -1, -1, -1, -1, -1, -1, -1};
 };
 
 struct GpuInfo {
@@ -81,6 +90,21 @@ struct GpuInfo {
     std::string uuid;            // unique id (hex) if available
     std::string pci_bdf;         // 0000:0a:00.0
     std::string drm_render;      // renderD128 style device name (attribution key)
+    // partitioning (MI3xx XCP/NPS): on a partitioned node each partition
+    // enumerates as its own device; partition_id says which slice of the
+    // physical GPU this device is. -1 = not reported.
+    std::string compute_partition;  // SPX/DPX/TPX/QPX/CPX, "" unknown
+    std::string memory_partition;   // NPS1/NPS2/NPS4/NPS8, "" unknown
+    int partition_id = -1;
+};
+
+// One probed counter family: available, or a reason why not (SURVEY.md §7:
+// per-counter availability must be probed with graceful degradation, not
+// assumed). Rendered as amd_counter_unavailable{counter,reason}.
+struct CounterProbe {
+    std::string counter;
+    bool available = false;
+    std::string reason;
 };
 
 class Backend {
@@ -90,14 +114,24 @@ class Backend {
     virtual GpuInfo info(int dev) = 0;
     virtual GpuSample sample(int dev) = 0;
     virtual std::string name() const = 0;
+    // Probe results for counters this backend cannot serve (and for ones
+    // it confirmed). Called once at startup; rendered as meta-metrics.
+    virtual std::vector<CounterProbe> probes() { return {}; }
 };
 
 // rsmi backend: returns nullptr (with err set) when librocm_smi64 is
 // unavailable or rsmi_init fails (no GPU).
 std::unique_ptr<Backend> make_rsmi_backend(std::string* err);
 
+// amd_smi backend: the successor library (rocm_smi_lib is in maintenance
+// mode upstream); nullptr + err when libamd_smi is unavailable or finds
+// no GPU. Same per-metric degradation contract as the rsmi backend.
+std::unique_ptr<Backend> make_amdsmi_backend(std::string* err);
+
 // mock backend: n synthetic devices; busy% follows a deterministic waveform
 // or the value in env MI355X_MOCK_BUSY / file `busy_file` when given.
+// Env MI355X_MOCK_PARTITIONS (1..8) makes each device report that many
+// compute partitions with per-XCP busy around the device busy%.
 std::unique_ptr<Backend> make_mock_backend(int n_devices, const std::string& busy_file);
 
 } // namespace mi355x

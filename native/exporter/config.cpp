@@ -16,6 +16,7 @@ const char* config_usage()
            "  -k, --kubernetes            enable kubelet pod attribution\n"
            "  --kubernetes-gpu-id-type <device-name|uuid|index>\n"
            "  --pod-resources-socket <path>\n"
+           "  --backend <auto|amdsmi|rsmi> counter library (default auto)\n"
            "  --mock <n>                  mock backend with n synthetic GPUs\n"
            "  --mock-busy-file <path>     scriptable busy%% for the mock backend\n"
            "  -v, --version\n"
@@ -128,6 +129,15 @@ bool parse_config(int argc, char** argv, Config* cfg, std::string* err)
         } else if (a == "--pod-resources-socket") {
             if (!(v = need(i))) goto missing;
             cfg->pod_resources_socket = v;
+            ++i;
+        } else if (a == "--backend") {
+            if (!(v = need(i))) goto missing;
+            cfg->backend = v;
+            if (cfg->backend != "auto" && cfg->backend != "amdsmi" &&
+                cfg->backend != "rsmi") {
+                if (err) *err = "bad --backend " + cfg->backend;
+                return false;
+            }
             ++i;
         } else if (a == "--mock") {
             if (!(v = need(i))) goto missing;

@@ -12,6 +12,9 @@
 // MI355X additions:
 //   --mock <n>          mock backend with n synthetic GPUs (CPU-only tests)
 //   --mock-busy-file <p> scriptable busy% source for the mock backend
+//   --backend <auto|amdsmi|rsmi>  counter library selection; auto prefers
+//                       amd-smi (rocm_smi_lib is in maintenance mode) and
+//                       falls back to rocm_smi
 //
 // Env fallbacks honored (so the reference's env-style config keeps working):
 //   DCGM_EXPORTER_LISTEN, DCGM_EXPORTER_KUBERNETES, DCGM_EXPORTER_INTERVAL,
@@ -36,6 +39,7 @@ struct Config {
     std::set<std::string> metric_set; // parsed from metric_file
     int mock_devices = 0;             // >0 => mock backend
     std::string mock_busy_file;
+    std::string backend = "auto";     // auto|amdsmi|rsmi
     bool show_help = false;
     bool show_version = false;
 };

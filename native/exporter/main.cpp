@@ -39,7 +39,7 @@ int main(int argc, char** argv)
         return 0;
     }
     if (cfg.show_version) {
-        std::puts("mi355x-exporter 0.1.0 (gfx950; rocm_smi backend)");
+        std::puts("mi355x-exporter 0.2.0 (gfx950; amd_smi/rocm_smi backends)");
         return 0;
     }
 
@@ -47,12 +47,20 @@ int main(int argc, char** argv)
     if (cfg.mock_devices > 0) {
         backend = make_mock_backend(cfg.mock_devices, cfg.mock_busy_file);
     } else {
-        backend = make_rsmi_backend(&err);
+        // auto: prefer amd-smi (rocm_smi_lib is in maintenance mode
+        // upstream), fall back to rocm_smi; --backend pins one.
+        std::string err_amdsmi, err_rsmi;
+        if (cfg.backend == "auto" || cfg.backend == "amdsmi")
+            backend = make_amdsmi_backend(&err_amdsmi);
+        if (!backend && (cfg.backend == "auto" || cfg.backend == "rsmi"))
+            backend = make_rsmi_backend(&err_rsmi);
         if (!backend) {
             std::fprintf(stderr,
-                         "mi355x-exporter: rocm_smi backend unavailable (%s); "
+                         "mi355x-exporter: no GPU backend available "
+                         "(amd_smi: %s; rocm_smi: %s); "
                          "use --mock N for a GPU-less stub\n",
-                         err.c_str());
+                         err_amdsmi.empty() ? "not tried" : err_amdsmi.c_str(),
+                         err_rsmi.empty() ? "not tried" : err_rsmi.c_str());
             return 3;
         }
     }
@@ -98,6 +106,11 @@ int main(int argc, char** argv)
     ropt.kubernetes = cfg.kubernetes;
     ropt.gpu_id_type = cfg.gpu_id_type;
     ropt.metric_set = cfg.metric_set;
+    ropt.probes = backend->probes();
+    for (auto& p : ropt.probes)
+        if (!p.available)
+            std::fprintf(stderr, "mi355x-exporter: counter %s unavailable: %s\n",
+                         p.counter.c_str(), p.reason.c_str());
 
     HttpServer server(
         cfg.listen_host, cfg.listen_port,

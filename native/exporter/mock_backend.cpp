@@ -56,6 +56,17 @@ class MockBackend final : public Backend {
         std::snprintf(b, sizeof(b), "0000:%02x:00.0", 0x10 + dev);
         gi.pci_bdf = b;
         gi.drm_render = "renderD" + std::to_string(128 + dev);
+        const char* np = std::getenv("MI355X_MOCK_PARTITIONS");
+        if (np && *np && std::atoi(np) > 0) {
+            int n_parts = std::atoi(np);
+            gi.compute_partition = n_parts >= 8   ? "CPX"
+                                   : n_parts == 4 ? "QPX"
+                                   : n_parts == 3 ? "TPX"
+                                   : n_parts == 2 ? "DPX"
+                                                  : "SPX";
+            gi.memory_partition = "NPS1";
+            gi.partition_id = 0;
+        }
         return gi;
     }
 
@@ -98,7 +109,35 @@ class MockBackend final : public Backend {
         s.ppt_residency_acc = elapsed * 0.10;    // 10% PVIOL
         s.thm_residency_acc = elapsed * 0.02;    // 2% TVIOL
         s.pcie_replay_count = 3;
+        // partitions: MI355X_MOCK_PARTITIONS=n (1..8) => per-XCP busy
+        // spread around the device busy% (partition p leans +/- p)
+        const char* np = std::getenv("MI355X_MOCK_PARTITIONS");
+        if (np && *np) {
+            int n_parts = std::atoi(np);
+            if (n_parts > kMaxXcp) n_parts = kMaxXcp;
+            for (int p = 0; p < n_parts; ++p) {
+                double b = s.busy_pct + (p - n_parts / 2.0);
+                if (b < 0) b = 0;
+                if (b > 100) b = 100;
+                s.xcp_busy_pct[p] = b;
+            }
+            s.num_partitions = n_parts > 0 ? n_parts : 0;
+        }
         return s;
+    }
+
+    std::vector<CounterProbe> probes() override
+    {
+        // mirror the real backends' probe surface so the renderer's
+        // meta-metric path is CPU-testable
+        std::vector<CounterProbe> out;
+        out.push_back({"mfma_activity", false,
+                       "no MFMA/matrix-pipe field in gpu_metrics v1.8; "
+                       "requires rocprofiler-sdk PMC (perturbs workloads)"});
+        const char* np = std::getenv("MI355X_MOCK_PARTITIONS");
+        out.push_back({"xcp_busy", np && *np,
+                       np && *np ? "" : "mock: MI355X_MOCK_PARTITIONS unset"});
+        return out;
     }
 
   private:

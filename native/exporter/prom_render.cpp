@@ -66,6 +66,19 @@ class Writer {
         out_ << "} " << buf << "\n";
     }
 
+    // family-filtered series with caller-provided labels only (no device)
+    void raw(const std::string& labels, double value)
+    {
+        if (!enabled(cur_family_)) return;
+        if (!cur_emitted_) {
+            out_ << cur_help_;
+            cur_emitted_ = true;
+        }
+        char buf[64];
+        std::snprintf(buf, sizeof(buf), "%.6g", value);
+        out_ << cur_family_ << "{" << labels << "} " << buf << "\n";
+    }
+
     std::string str() const { return out_.str(); }
 
   private:
@@ -224,11 +237,57 @@ std::string render_metrics(const std::vector<DeviceMetrics>& devs,
     w.family("amd_gpu_hotspot_temp", "Junction/hotspot temperature (C).", "gauge");
     for (auto& d : devs) w.sample(d, attr, d.sample.temp_hotspot_c);
 
+    w.family("amd_xcp_busy_percent",
+             "Per-compute-partition (XCP) instantaneous gfx busy % - mean "
+             "over the partition's XCCs (gpu_metrics xcp_stats).",
+             "gauge");
+    for (auto& d : devs)
+        for (int p = 0; p < d.sample.num_partitions && p < kMaxXcp; ++p)
+            w.sample(d, attr, d.sample.xcp_busy_pct[p],
+                     "partition=\"" + std::to_string(p) + "\"");
+
+    w.family("amd_compute_partition_info",
+             "Compute/memory partition mode of this device (value is "
+             "always 1; the information is in the labels).",
+             "gauge");
+    for (auto& d : devs) {
+        if (d.info.compute_partition.empty() &&
+            d.info.memory_partition.empty() && d.info.partition_id < 0)
+            continue;
+        std::string extra;
+        if (!d.info.compute_partition.empty())
+            extra += "compute=\"" + d.info.compute_partition + "\"";
+        if (!d.info.memory_partition.empty()) {
+            if (!extra.empty()) extra += ",";
+            extra += "memory=\"" + d.info.memory_partition + "\"";
+        }
+        if (d.info.partition_id >= 0) {
+            if (!extra.empty()) extra += ",";
+            extra += "partition_id=\"" + std::to_string(d.info.partition_id) +
+                     "\"";
+        }
+        w.sample(d, attr, 1.0, extra);
+    }
+
     w.family("amd_exporter_samples_total",
              "Sampling ticks taken for this device since exporter start "
              "(freshness counter: advances once per -c interval).",
              "counter");
     for (auto& d : devs) w.sample(d, attr, (double)d.samples_taken);
+
+    // counter-availability meta-metrics (probe, don't
+    // assume): one series per counter the backend cannot serve, with the
+    // reason as a label — what "graceful degradation" looks like on the
+    // wire. Emitted without device labels (the probe is backend-global).
+    w.family("amd_counter_unavailable",
+             "A probed counter family this exporter cannot serve on this "
+             "node, with the reason (value always 1).",
+             "gauge");
+    for (auto& p : opt.probes)
+        if (!p.available)
+            w.raw("counter=\"" + esc(p.counter) + "\",reason=\"" +
+                      esc(p.reason) + "\"",
+                  1.0);
 
     return w.str();
 }

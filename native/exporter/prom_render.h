@@ -15,6 +15,7 @@
 
 #pragma once
 
+#include "backend.h"
 #include "sampler.h"
 
 #include <map>
@@ -42,6 +43,9 @@ struct RenderOptions {
     // empty set = all metrics; otherwise only families named here (the
     // reference's `-f` metric-set file, one name per line, '#' comments).
     std::set<std::string> metric_set;
+    // counter-availability probe results (Backend::probes()), rendered as
+    // amd_counter_unavailable meta-metrics so degradation is observable.
+    std::vector<CounterProbe> probes;
 };
 
 std::string attribution_key(const GpuInfo& info, const std::string& id_type);

@@ -11,6 +11,7 @@
 // than crashing the daemon.
 
 #include "backend.h"
+#include "gpu_metrics_parse.h"
 
 #include <dlfcn.h>
 #include <rocm_smi/rocm_smi.h>
@@ -50,6 +51,11 @@ struct RsmiApi {
     DECL(rsmi_dev_gpu_clk_freq_get);
     DECL(rsmi_dev_gpu_metrics_info_get);
     DECL(rsmi_dev_ecc_count_get);
+    // partition APIs (optional: absent in very old librocm_smi64 builds;
+    // resolved lazily, null => per-metric degradation)
+    DECL(rsmi_dev_compute_partition_get);
+    DECL(rsmi_dev_memory_partition_get);
+    DECL(rsmi_dev_partition_id_get);
 #undef DECL
 
     bool load(std::string* err)
@@ -88,6 +94,16 @@ struct RsmiApi {
         RESOLVE(rsmi_dev_gpu_metrics_info_get)
         RESOLVE(rsmi_dev_ecc_count_get)
 #undef RESOLVE
+        // optional symbols: no failure when missing
+        rsmi_dev_compute_partition_get =
+            reinterpret_cast<decltype(&::rsmi_dev_compute_partition_get)>(
+                dlsym(handle, "rsmi_dev_compute_partition_get"));
+        rsmi_dev_memory_partition_get =
+            reinterpret_cast<decltype(&::rsmi_dev_memory_partition_get)>(
+                dlsym(handle, "rsmi_dev_memory_partition_get"));
+        rsmi_dev_partition_id_get =
+            reinterpret_cast<decltype(&::rsmi_dev_partition_id_get)>(
+                dlsym(handle, "rsmi_dev_partition_id_get"));
         return true;
     }
 };
@@ -127,7 +143,57 @@ class RsmiBackend final : public Backend {
         if (api_.rsmi_dev_drm_render_minor_get(dev, &minor) == RSMI_STATUS_SUCCESS) {
             gi.drm_render = "renderD" + std::to_string(minor);
         }
+        char pbuf[32] = {0};
+        if (api_.rsmi_dev_compute_partition_get &&
+            api_.rsmi_dev_compute_partition_get(dev, pbuf, sizeof(pbuf)) ==
+                RSMI_STATUS_SUCCESS)
+            gi.compute_partition = pbuf;
+        std::memset(pbuf, 0, sizeof(pbuf));
+        if (api_.rsmi_dev_memory_partition_get &&
+            api_.rsmi_dev_memory_partition_get(dev, pbuf, sizeof(pbuf)) ==
+                RSMI_STATUS_SUCCESS)
+            gi.memory_partition = pbuf;
+        uint32_t pid = 0;
+        if (api_.rsmi_dev_partition_id_get &&
+            api_.rsmi_dev_partition_id_get(dev, &pid) == RSMI_STATUS_SUCCESS &&
+            pid != 0xffffffffu)
+            gi.partition_id = (int)pid;
         return gi;
+    }
+
+    std::vector<CounterProbe> probes() override
+    {
+        std::vector<CounterProbe> out;
+        // MFMA / matrix-pipe activity (BASELINE.json north star names it):
+        // probed against the full gpu_metrics layout — no MFMA/matrix-pipe
+        // field exists through gpu_metrics v1.8 (ROCm 7.2 headers); the
+        // only source is rocprofiler-sdk PMC sampling, which perturbs
+        // co-running workloads and conflicts with attached profilers
+        // (docs/METRICS.md), so the counter degrades to "unavailable"
+        // rather than shipping a perturbing default-on sampler.
+        out.push_back({"mfma_activity", false,
+                       "no MFMA/matrix-pipe field in gpu_metrics v1.8; "
+                       "requires rocprofiler-sdk PMC (perturbs workloads)"});
+        rsmi_gpu_metrics_t gm;
+        std::memset(&gm, 0, sizeof(gm));
+        bool gm_ok = n_ > 0 && api_.rsmi_dev_gpu_metrics_info_get(0, &gm) ==
+                                   RSMI_STATUS_SUCCESS;
+        if (!gm_ok) {
+            out.push_back({"xcp_busy", false, "gpu_metrics read failed"});
+        } else if (gm.num_partition == 0xffff) {
+            out.push_back({"xcp_busy", false,
+                           "gpu_metrics reports no partition count "
+                           "(pre-v1.6 firmware)"});
+        } else {
+            out.push_back({"xcp_busy", true, ""});
+        }
+        if (!api_.rsmi_dev_compute_partition_get)
+            out.push_back({"compute_partition", false,
+                           "librocm_smi64 lacks "
+                           "rsmi_dev_compute_partition_get"});
+        else
+            out.push_back({"compute_partition", true, ""});
+        return out;
     }
 
     GpuSample sample(int dev) override
@@ -209,65 +275,9 @@ class RsmiBackend final : public Backend {
 
         rsmi_gpu_metrics_t gm;
         std::memset(&gm, 0, sizeof(gm));
-        if (api_.rsmi_dev_gpu_metrics_info_get(dev, &gm) == RSMI_STATUS_SUCCESS) {
-            auto u16ok = [](uint16_t x) { return x != 0xffff; };
-            if (u16ok(gm.average_gfx_activity)) {
-                s.gfx_activity_pct = gm.average_gfx_activity;
-                if (s.busy_pct < 0) s.busy_pct = gm.average_gfx_activity;
-                s.ok = true;
-            }
-            if (u16ok(gm.average_umc_activity)) {
-                s.umc_activity_pct = gm.average_umc_activity;
-                if (s.mem_busy_pct < 0) s.mem_busy_pct = gm.average_umc_activity;
-            }
-            if (gm.gfx_activity_acc != 0xffffffffu)
-                s.gfx_activity_acc = gm.gfx_activity_acc;
-            if (gm.mem_activity_acc != 0xffffffffu)
-                s.mem_activity_acc = gm.mem_activity_acc;
-            if (u16ok(gm.current_socket_power) && s.power_w < 0)
-                s.power_w = gm.current_socket_power;
-            if (u16ok(gm.current_gfxclk) && s.sclk_mhz < 0)
-                s.sclk_mhz = gm.current_gfxclk;
-            if (u16ok(gm.current_uclk) && s.mclk_mhz < 0)
-                s.mclk_mhz = gm.current_uclk;
-            if (u16ok(gm.temperature_mem) && s.temp_mem_c < 0)
-                s.temp_mem_c = gm.temperature_mem;
-            if (u16ok(gm.temperature_edge) && gm.temperature_edge != 0 &&
-                s.temp_edge_c < 0)
-                s.temp_edge_c = gm.temperature_edge;
-            if (u16ok(gm.temperature_hotspot) && s.temp_hotspot_c < 0)
-                s.temp_hotspot_c = gm.temperature_hotspot;
+        if (api_.rsmi_dev_gpu_metrics_info_get(dev, &gm) == RSMI_STATUS_SUCCESS)
+            parse_gpu_metrics(gm, s);
 
-            // PCIe: accumulated GB/s-seconds; expose instantaneous if present
-            if (gm.pcie_bandwidth_inst && gm.pcie_bandwidth_inst != ~0ull) {
-                s.pcie_tx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
-                s.pcie_rx_bps = gm.pcie_bandwidth_inst * 1e9 / 2.0;
-            }
-
-            if (gm.accumulation_counter && gm.accumulation_counter != ~0ull) {
-                s.accumulation_counter = (double)gm.accumulation_counter;
-                if (gm.ppt_residency_acc != ~0ull)
-                    s.ppt_residency_acc = (double)gm.ppt_residency_acc;
-                if (gm.socket_thm_residency_acc != ~0ull)
-                    s.thm_residency_acc = (double)gm.socket_thm_residency_acc;
-            }
-            if (gm.pcie_replay_count_acc != ~0ull)
-                s.pcie_replay_count = (double)gm.pcie_replay_count_acc;
-
-            if (u16ok(gm.xgmi_link_width)) s.xgmi_link_width = gm.xgmi_link_width;
-            if (u16ok(gm.xgmi_link_speed)) s.xgmi_link_speed_gbps = gm.xgmi_link_speed;
-            int nl = 0;
-            for (int i = 0; i < kMaxXgmiLinks; ++i) {
-                uint64_t r = gm.xgmi_read_data_acc[i];
-                uint64_t w = gm.xgmi_write_data_acc[i];
-                if (r == ~0ull) r = 0;
-                if (w == ~0ull) w = 0;
-                s.xgmi_read_acc_kb[i] = (double)r;
-                s.xgmi_write_acc_kb[i] = (double)w;
-                if (gm.xgmi_link_status[i] == 1 || r || w) nl = i + 1;
-            }
-            s.xgmi_num_links = nl;
-        }
         // dcgm_gpu_temp must exist (README.md:46 probe): some MI3xx SKUs
         // expose no edge sensor — fall back to junction/hotspot.
         if (s.temp_edge_c < 0 && s.temp_hotspot_c >= 0)

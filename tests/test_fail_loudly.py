@@ -37,4 +37,4 @@ def test_exporter_refuses_to_fake_gpus_without_backend():
     r = subprocess.run([EXPORTER_BIN, "-c", "100", "-l", "127.0.0.1:0"],
                        capture_output=True, timeout=30)
     assert r.returncode == 3
-    assert b"backend unavailable" in r.stderr
+    assert b"no GPU backend available" in r.stderr
