@@ -78,6 +78,20 @@ def _load():
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_double, ctypes.c_void_p,
         ]
+        lib.lg_gemm_fp8_bench.argtypes = [
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double),
+        ]
+        lib.lg_gemm_fp8_verify.argtypes = [
+            ctypes.c_int,
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ]
         lib.lg_bw_burn.argtypes = [
             ctypes.c_int, ctypes.c_double, ctypes.c_double, ctypes.c_double,
             ctypes.c_double, ctypes.c_void_p, ctypes.POINTER(ctypes.c_double),
@@ -140,6 +154,34 @@ def gemm_bench(m=4096, n=4096, k=4096, warmup=5, iters=50, device=0,
     _check(_load().lg_gemm_bf16_bench_variant(device, m, n, k, warmup, iters,
                                               variant,
                                               ctypes.byref(ms), ctypes.byref(tf)))
+    return ms.value, tf.value
+
+
+def gemm_fp8(a: np.ndarray, bt: np.ndarray, device: int = 0):
+    """FP8 (E4M3) MFMA GEMM numerics entry: quantizes the f32 inputs to
+    E4M3, computes C = Aq @ Btq^T on the GPU, and returns
+    (c, aq, btq) where aq/btq are the dequantized (f32) operands the GPU
+    actually multiplied — the caller computes the exact reference from
+    them. M,N %% 256 == 0, K %% 128 == 0."""
+    m, k = a.shape
+    n, k2 = bt.shape
+    assert k == k2
+    a = np.ascontiguousarray(a, np.float32)
+    bt = np.ascontiguousarray(bt, np.float32)
+    c = np.empty((m, n), np.float32)
+    aq = np.empty_like(a)
+    btq = np.empty_like(bt)
+    _check(_load().lg_gemm_fp8_verify(device, a, bt, c, aq, btq, m, n, k))
+    return c, aq, btq
+
+
+def gemm_fp8_bench(m=8192, n=8192, k=8192, warmup=2, iters=10, raster=1,
+                   device=0):
+    """Returns (ms_per_gemm, tflops) for the fp8 kernel."""
+    ms = ctypes.c_double()
+    tf = ctypes.c_double()
+    _check(_load().lg_gemm_fp8_bench(device, m, n, k, warmup, iters, raster,
+                                     ctypes.byref(ms), ctypes.byref(tf)))
     return ms.value, tf.value
 
 

@@ -1,0 +1,219 @@
+// gemm_fp8_256.hip — 256x256-tile FP8 (E4M3) MFMA GEMM for MI355X (gfx950).
+//
+// CDNA4's fp8 dense peak is 2x bf16 (~5 PF/s vs ~2.5 PF/s): the matrix
+// core's native fp8 shape is v_mfma_scale_f32_16x16x128_f8f6f4 — K=128 per
+// instruction, 32 fp8 operand bytes per lane, f32 accumulate, with an
+// E8M0 block-scale operand (we pass the 127 bias = x1.0; this is a plain
+// fp8 GEMM, not MXFP8).
+//
+// The schedule is the d9 single-barrier-per-K-tile design proven on the
+// bf16 kernel (gemm_bf16_256.hip, profiles/gemm_bf16_256_ladder.md),
+// byte-for-byte compatible in its LDS machinery: an fp8 half-tile image is
+// [128 rows][128 k] x 1 B = 16 KiB with 128-B rows — the same image size,
+// row size, st_16x32 swizzle and 2-glds-per-wave staging as the bf16
+// [128][64] x 2 B images, while each K-tile now covers K=128 (double the
+// FLOPs for the same LDS traffic: 24 ds_read_b128 per wave per tile).
+//
+//   per K-tile (K=128), phase q of 4: m-frags {2q,2q+1} x all 4 n-frags
+//     reads:  A 2 frags x 2xb128; B (q0 only) 4 frags x 2xb128, held
+//     stages: q0 A0,A1 -> buf^1; q1 B0; q2 B1  (all slots dead since the
+//             previous boundary barrier — single barrier per tile)
+//     MFMA:   8 x mfma_scale_f32_16x16x128_f8f6f4 (m2 x n4)
+//
+// C[M][N] f32 = A[M][K] fp8 @ B^T[N][K] fp8; M,N % 256 == 0, K % 128 == 0.
+
+#include <hip/hip_runtime.h>
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) int i32x4;
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+
+#define F8_HALF_BYTES 16384   // [128][128] fp8 image
+
+static __device__ __forceinline__ int swz256b(int byte_off)
+{
+    return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_fp8_tn_256_impl(
+    const unsigned char* __restrict__ A,   // [M][K] fp8 E4M3
+    const unsigned char* __restrict__ Bt,  // [N][K] fp8 E4M3
+    float* __restrict__ C,                 // [M][N] f32
+    int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned char lds[2 * 4 * F8_HALF_BYTES];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;  // 0..1: A half this wave consumes
+    const int wc = w & 3;   // 0..3: B cols wc*64..+64
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 128;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    // glds source mapping (piece p = w*2+it covers image rows p*8..p*8+7;
+    // 128-B rows, byte-granular k)
+    const int in_piece = swz256b(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kb = in_piece & 127;   // byte (= k index) within the row
+
+    // fragment read byte offset: row `row_in_half`, 32-B k-chunk per lane
+    // group (lane>>4), `half16` selects the chunk's low/high 16 B
+    auto frag_off = [&](int row_in_half, int half16) {
+        return swz256b(row_in_half * 128 + (lane >> 4) * 32 + half16 * 16);
+    };
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;  // tail clamp (benign restage)
+            const long k0 = (long)kt * 128 + src_kb;
+            const unsigned char* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned char* dst = &lds[(buf * 4 + h) * F8_HALF_BYTES];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 1024),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        i32x8 afrag[2];
+        i32x8 bfrag[4];
+
+        auto read_frag = [&](const unsigned char* base, int row_in_half) {
+            i32x4 lo = *(const i32x4*)(base + frag_off(row_in_half, 0));
+            i32x4 hi = *(const i32x4*)(base + frag_off(row_in_half, 1));
+            i32x8 r;
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                r[e] = lo[e];
+                r[e + 4] = hi[e];
+            }
+            return r;
+        };
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const unsigned char* la = &lds[(buf * 4 + wr) * F8_HALF_BYTES];
+            const unsigned char* lb =
+                &lds[(buf * 4 + 2 + (wc >> 1)) * F8_HALF_BYTES];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = q * 2;
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+                    afrag[m] =
+                        read_frag(la, (mbase + m) * 16 + (lane & 15));
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+                        bfrag[n] =
+                            read_frag(lb, bcol0 + n * 16 + (lane & 15));
+                }
+
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 1, 2, buf ^ 1);
+                } else if (q == 2) {
+                    stage(kt + 1, 3, buf ^ 1);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+                        // cbsz=0/blgp=0: both operands FP8 (E4M3);
+                        // scale bytes 0x7F = E8M0 bias = x1.0 (plain fp8)
+                        acc[mbase + m][n] =
+                            __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                                afrag[m], bfrag[n], acc[mbase + m][n],
+                                0, 0, 0, 0x7F, 0, 0x7F);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_fp8_tn_256(
+    const unsigned char* A, const unsigned char* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_fp8_tn_256_impl<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// raster ablation
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_fp8_tn_256_nr(
+    const unsigned char* A, const unsigned char* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_fp8_tn_256_impl<0>(A, Bt, C, M, N, K, tiles_per_cta);
+}

@@ -19,6 +19,7 @@
 #include <hip/hip_runtime.h>
 
 #include <chrono>
+#include <cmath>
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
@@ -51,6 +52,12 @@ extern "C" __global__ void gemm_bf16_tn_256_d1(const unsigned short*, const unsi
 extern "C" __global__ void gemm_bf16_tn_256(const unsigned short*,
                                             const unsigned short*, float*, int,
                                             int, int, int);
+extern "C" __global__ void gemm_fp8_tn_256(const unsigned char*,
+                                           const unsigned char*, float*, int,
+                                           int, int, int);
+extern "C" __global__ void gemm_fp8_tn_256_nr(const unsigned char*,
+                                              const unsigned char*, float*, int,
+                                              int, int, int);
 
 #define LG_CHECK(expr)                                                        \
     do {                                                                      \
@@ -443,3 +450,159 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
 }
 
 } // extern "C"
+
+// ---------------------------------------------------------------------------
+// FP8 (E4M3) MFMA GEMM — the CDNA4 2x-peak datatype (gemm_fp8_256.hip)
+// ---------------------------------------------------------------------------
+
+// f32 -> OCP E4M3 (saturating, round-to-nearest-even via the FPU's default
+// rounding). Max finite 448; subnormal step 2^-9.
+static unsigned char f32_to_fp8_e4m3(float f)
+{
+    if (f != f) return 0x7f;  // NaN
+    unsigned char sign = f < 0 ? 0x80 : 0;
+    float af = f < 0 ? -f : f;
+    if (af > 448.f) return sign | 0x7e;  // saturate to max finite
+    int ef;
+    float m = std::frexp(af, &ef);  // af = m * 2^ef, m in [0.5, 1)
+    int k = ef - 1;                 // af = (2m) * 2^k, 2m in [1, 2)
+    if (af == 0.f || k < -9) return sign;
+    if (k < -6) {  // subnormal band: units of 2^-9, codes 1..8
+        int q = (int)std::nearbyint(af * 512.f);
+        if (q == 0) return sign;
+        return (unsigned char)(sign | q);  // q==8 lands on min normal 0x08
+    }
+    int q = (int)std::nearbyint(std::ldexp(af, 3 - k));  // in [8, 16]
+    if (q == 16) {
+        ++k;
+        q = 8;
+        if (k > 8) return sign | 0x7e;
+    }
+    return (unsigned char)(sign | ((k + 7) << 3) | (q - 8));
+}
+
+static float fp8_e4m3_to_f32(unsigned char v)
+{
+    int sign = v >> 7;
+    int exp = (v >> 3) & 0xf;
+    int man = v & 7;
+    float r;
+    if (exp == 0)
+        r = man * (1.f / 512.f);
+    else if (exp == 15 && man == 7)
+        r = __builtin_nanf("");
+    else
+        r = (1.f + man / 8.f) * std::ldexp(1.f, exp - 7);
+    return sign ? -r : r;
+}
+
+struct Fp8GemmBufs {
+    unsigned char* a = nullptr;
+    unsigned char* bt = nullptr;
+    float* c = nullptr;
+    int m = 0, n = 0, k = 0;
+};
+
+static int fp8_gemm_alloc(Fp8GemmBufs& g, int m, int n, int k, bool fill_random)
+{
+    LG_CHECK(hipMalloc(&g.a, (size_t)m * k));
+    LG_CHECK(hipMalloc(&g.bt, (size_t)n * k));
+    LG_CHECK(hipMalloc(&g.c, (size_t)m * n * 4));
+    g.m = m; g.n = n; g.k = k;
+    if (fill_random) {
+        // random fp8 in [-1, 1): DVFS-honest load (playbook rule 25)
+        size_t na = (size_t)m * k, nb = (size_t)n * k;
+        std::vector<unsigned char> h(na > nb ? na : nb);
+        uint32_t st = 0x2468ace1u;
+        for (size_t i = 0; i < h.size(); ++i) {
+            st = st * 1664525u + 1013904223u;
+            float f = ((st >> 8) & 0xffff) / 32768.0f - 1.0f;
+            h[i] = f32_to_fp8_e4m3(f);
+        }
+        LG_CHECK(hipMemcpy(g.a, h.data(), na, hipMemcpyHostToDevice));
+        LG_CHECK(hipMemcpy(g.bt, h.data(), nb, hipMemcpyHostToDevice));
+    }
+    return 0;
+}
+
+static void fp8_gemm_free(Fp8GemmBufs& g)
+{
+    (void)hipFree(g.a); (void)hipFree(g.bt); (void)hipFree(g.c);
+    g = Fp8GemmBufs{};
+}
+
+static int fp8_gemm_launch(const Fp8GemmBufs& g, hipStream_t stream, int raster)
+{
+    int n_tiles = (g.m / 256) * (g.n / 256);
+    int blocks = n_tiles < 2048 ? n_tiles : 2048;
+    int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
+    hipLaunchKernelGGL(raster ? gemm_fp8_tn_256 : gemm_fp8_tn_256_nr,
+                       dim3(blocks), dim3(512), 0, stream, g.a, g.bt, g.c,
+                       g.m, g.n, g.k, tiles_per_cta);
+    return 0;
+}
+
+extern "C" {
+
+// Timed fp8 GEMM: mean ms + TF/s over `iters` launches. raster: 4x4
+// super-tile rasterization on (1, the product config) or off (0).
+int lg_gemm_fp8_bench(int device, int m, int n, int k, int warmup, int iters,
+                      int raster, double* ms_out, double* tflops_out)
+{
+    if (m % 256 || n % 256 || k % 128) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "fp8 gemm dims must be multiples of 256/256/128");
+        return -1;
+    }
+    LG_CHECK(hipSetDevice(device));
+    Fp8GemmBufs g;
+    if (fp8_gemm_alloc(g, m, n, k, true)) return -1;
+    for (int i = 0; i < warmup; ++i) fp8_gemm_launch(g, 0, raster);
+    LG_CHECK(hipDeviceSynchronize());
+    double t0 = now_ms();
+    for (int i = 0; i < iters; ++i) fp8_gemm_launch(g, 0, raster);
+    LG_CHECK(hipDeviceSynchronize());
+    double t1 = now_ms();
+    double ms = (t1 - t0) / iters;
+    if (ms_out) *ms_out = ms;
+    if (tflops_out) *tflops_out = 2.0 * m * n * k / (ms * 1e-3) / 1e12;
+    fp8_gemm_free(g);
+    return 0;
+}
+
+// Numerics entry: quantizes the f32 inputs to E4M3, runs the kernel, and
+// writes C plus the DEQUANTIZED inputs (aq/btq, f32) back so the caller
+// can compute the exact reference matmul on what the GPU actually saw.
+int lg_gemm_fp8_verify(int device, const float* a_h, const float* bt_h,
+                       float* c_out, float* aq_out, float* btq_out,
+                       int m, int n, int k)
+{
+    if (m % 256 || n % 256 || k % 128) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "fp8 gemm dims must be multiples of 256/256/128");
+        return -1;
+    }
+    LG_CHECK(hipSetDevice(device));
+    Fp8GemmBufs g;
+    if (fp8_gemm_alloc(g, m, n, k, false)) return -1;
+    std::vector<unsigned char> q((size_t)m * k);
+    for (size_t i = 0; i < q.size(); ++i) {
+        q[i] = f32_to_fp8_e4m3(a_h[i]);
+        if (aq_out) aq_out[i] = fp8_e4m3_to_f32(q[i]);
+    }
+    LG_CHECK(hipMemcpy(g.a, q.data(), q.size(), hipMemcpyHostToDevice));
+    q.resize((size_t)n * k);
+    for (size_t i = 0; i < q.size(); ++i) {
+        q[i] = f32_to_fp8_e4m3(bt_h[i]);
+        if (btq_out) btq_out[i] = fp8_e4m3_to_f32(q[i]);
+    }
+    LG_CHECK(hipMemcpy(g.bt, q.data(), q.size(), hipMemcpyHostToDevice));
+    fp8_gemm_launch(g, 0, 1);
+    LG_CHECK(hipDeviceSynchronize());
+    LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
+    fp8_gemm_free(g);
+    return 0;
+}
+
+} // extern "C"
+

@@ -6,7 +6,8 @@ Runs each variant's bench entry several times interleaved (A,B,A,B,...) so
 DVFS drift hits all variants equally; prints per-run TF/s and the max.
 Variant map: 2=product d6+raster, 7=d7 5-barrier, 8=d6 no-raster,
 10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster, 13=d14
-16-wave 4-waves/SIMD, 14=d6 round-1 product (now ablation).
+16-wave 4-waves/SIMD, 14=d6 round-1 product (now ablation),
+100/101=fp8 E4M3 kernel with/without raster.
 """
 
 import json
@@ -40,8 +41,13 @@ def main():
         res = {v: [] for v in variants}
         for r in range(rounds):
             for v in variants:
-                ms, tf = loadgen.gemm_bench(m, n, k, warmup=2, iters=4,
-                                            variant=v)
+                if v >= 100:  # 100/101: fp8 kernel raster on/off
+                    ms, tf = loadgen.gemm_fp8_bench(m, n, k, warmup=2,
+                                                    iters=4,
+                                                    raster=(v == 100))
+                else:
+                    ms, tf = loadgen.gemm_bench(m, n, k, warmup=2, iters=4,
+                                                variant=v)
                 res[v].append(round(tf, 1))
                 print(f"{sh} v{v} round{r}: {tf:.0f} TF/s", file=sys.stderr)
         out[sh] = {str(v): {"tf_runs": res[v], "tf_max": max(res[v])}
