@@ -466,7 +466,7 @@ static unsigned char f32_to_fp8_e4m3(float f)
     int ef;
     float m = std::frexp(af, &ef);  // af = m * 2^ef, m in [0.5, 1)
     int k = ef - 1;                 // af = (2m) * 2^k, 2m in [1, 2)
-    if (af == 0.f || k < -9) return sign;
+    if (af == 0.f || k < -20) return sign;  // below half the min subnormal
     if (k < -6) {  // subnormal band: units of 2^-9, codes 1..8
         int q = (int)std::nearbyint(af * 512.f);
         if (q == 0) return sign;
