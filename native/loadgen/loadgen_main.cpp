@@ -27,6 +27,8 @@ int lg_gemm_bf16_bench(int device, int m, int n, int k, int warmup, int iters,
 int lg_gemm_bf16_bench_variant(int device, int m, int n, int k, int warmup,
                                int iters, int variant, double* ms_out,
                                double* tflops_out);
+int lg_gemm_fp8_bench(int device, int m, int n, int k, int warmup, int iters,
+                      int raster, double* ms_out, double* tflops_out);
 int lg_gemm_burn(int device, double target_util_pct, double seconds,
                  int m, int n, int k, double period_ms, volatile int* stop_flag);
 int lg_bw_burn(int device, double target_util_pct, double seconds, double gb,
@@ -44,7 +46,7 @@ int main(int argc, char** argv)
 {
     if (argc < 2) {
         std::fprintf(stderr,
-                     "usage: mi355x-loadgen {vectoradd|gemm|burn|bwburn|devices} [flags]\n");
+                     "usage: mi355x-loadgen {vectoradd|gemm|fp8gemm|burn|bwburn|devices} [flags]\n");
         return 1;
     }
     std::string mode = argv[1];
@@ -90,6 +92,26 @@ int main(int argc, char** argv)
         else if (variant >= 2) std::snprintf(vname, sizeof(vname), "_v%d", variant);
         std::printf("gemm_bf16%s %dx%dx%d ms=%.3f tflops=%.1f\n",
                     vname, m, n, k, ms, tf);
+        return 0;
+    }
+    if (mode == "fp8gemm") {
+        int m = (int)argd(argc, argv, "--m", 8192);
+        int n = (int)argd(argc, argv, "--n-dim", 8192);
+        int k = (int)argd(argc, argv, "--k", 8192);
+        int iters = (int)argd(argc, argv, "--iters", 10);
+        int warmup = (int)argd(argc, argv, "--warmup", 2);
+        int raster = 1;
+        for (int i = 1; i < argc; ++i)
+            if (!std::strcmp(argv[i], "--no-raster")) raster = 0;
+        double ms = 0, tf = 0;
+        if (lg_gemm_fp8_bench(device, m, n, k, warmup, iters, raster, &ms,
+                              &tf)) {
+            std::fprintf(stderr, "error: %s
+", lg_last_error());
+            return 2;
+        }
+        std::printf("gemm_fp8 %dx%dx%d ms=%.3f tflops=%.1f
+", m, n, k, ms, tf);
         return 0;
     }
     if (mode == "burn") {
