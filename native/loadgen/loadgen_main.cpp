@@ -106,12 +106,10 @@ int main(int argc, char** argv)
         double ms = 0, tf = 0;
         if (lg_gemm_fp8_bench(device, m, n, k, warmup, iters, raster, &ms,
                               &tf)) {
-            std::fprintf(stderr, "error: %s
-", lg_last_error());
+            std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
-        std::printf("gemm_fp8 %dx%dx%d ms=%.3f tflops=%.1f
-", m, n, k, ms, tf);
+        std::printf("gemm_fp8 %dx%dx%d ms=%.3f tflops=%.1f\n", m, n, k, ms, tf);
         return 0;
     }
     if (mode == "burn") {
