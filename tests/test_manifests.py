@@ -192,3 +192,33 @@ class TestKustomize:
         for res in k["resources"]:
             if res != "../":
                 assert (DEPLOY / "multi-metric" / res).exists(), res
+
+
+class TestKindHarnessScript:
+    """deploy/kind/run.sh is the executable form of the runbook: its
+    referenced manifests must exist and its probes must match the
+    reference's verification steps (it cannot run here — no Docker — but
+    it must not drift from the files it applies)."""
+
+    def setup_method(self):
+        self.script = (DEPLOY / "kind" / "run.sh").read_text()
+
+    def test_applied_files_exist(self):
+        import re
+        for m in re.finditer(r"apply -f (\S+)", self.script):
+            path = (DEPLOY / "kind" / m.group(1)).resolve()
+            assert path.exists(), m.group(1)
+
+    def test_values_file_referenced(self):
+        assert "../kube-prometheus-stack-values.yaml" in self.script
+
+    def test_reference_probes_present(self):
+        # the four reference verification seams (SURVEY.md §4)
+        assert "dcgm_gpu_temp" in self.script              # README.md:42-47
+        assert "cuda_test_gpu_avg" in self.script          # README.md:80-88
+        assert "custom.metrics.k8s.io" in self.script      # README.md:98-102
+        assert "hpa_scaled_up" in self.script              # README.md:112-122
+
+    def test_fails_loudly(self):
+        assert "set -euo pipefail" in self.script
+        assert "fail()" in self.script
