@@ -99,6 +99,7 @@ def main():
                              downscale_stabilization_s=30.0),
             extra_samples=lambda: synth_pod_labels(
                 ["cuda-test-0"]),
+            use_adapter=True,  # full L4 hop (adapter discovery + GET)
         )
 
         t_start = time.monotonic()
