@@ -86,6 +86,8 @@ bool parse_config(int argc, char** argv, Config* cfg, std::string* err)
         cfg->kubernetes = truthy(v);
     if (const char* v = std::getenv("MI355X_EXPORTER_KUBERNETES"))
         cfg->kubernetes = truthy(v);
+    if (const char* v = std::getenv("MI355X_EXPORTER_BACKEND"))
+        cfg->backend = v;
 
     auto need = [&](int i) -> const char* {
         if (i + 1 >= argc) return nullptr;
@@ -154,6 +156,11 @@ bool parse_config(int argc, char** argv, Config* cfg, std::string* err)
         continue;
     missing:
         if (err) *err = "flag " + a + " needs a value";
+        return false;
+    }
+    if (cfg->backend != "auto" && cfg->backend != "amdsmi" &&
+        cfg->backend != "rsmi") {
+        if (err) *err = "bad backend " + cfg->backend;
         return false;
     }
     if (cfg->interval_ms < 10) {

@@ -269,6 +269,12 @@ std::string render_metrics(const std::vector<DeviceMetrics>& devs,
         w.sample(d, attr, 1.0, extra);
     }
 
+    w.family("amd_exporter_sample_duration_ms",
+             "Wall time of the last per-device sampling pass (ms) — the "
+             "exporter's own overhead, for observer-effect monitoring.",
+             "gauge");
+    for (auto& d : devs) w.sample(d, attr, d.sample_pass_ms);
+
     w.family("amd_exporter_samples_total",
              "Sampling ticks taken for this device since exporter start "
              "(freshness counter: advances once per -c interval).",

@@ -31,7 +31,11 @@ void Sampler::sample_once()
 {
     int n = backend_->device_count();
     std::vector<GpuSample> fresh(n);
+    auto t0 = std::chrono::steady_clock::now();
     for (int i = 0; i < n; ++i) fresh[i] = backend_->sample(i);
+    double pass_ms = std::chrono::duration<double, std::milli>(
+                         std::chrono::steady_clock::now() - t0)
+                         .count() / (n > 0 ? n : 1);
 
     std::lock_guard<std::mutex> lk(mu_);
     bool any_ok = false;
@@ -87,6 +91,7 @@ void Sampler::sample_once()
             }
         }
         d.sample = cur;
+        d.sample_pass_ms = pass_ms;
         d.samples_taken++;
     }
     if (any_ok) ready_ = true;

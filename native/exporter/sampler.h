@@ -27,6 +27,7 @@ struct DeviceMetrics {
     double xgmi_total_bps = 0;
     double busy_windowed_pct = -1; // from gfx_activity_acc delta when available
     double mem_busy_windowed_pct = -1;
+    double sample_pass_ms = -1;    // wall time of the last sampling pass
     double pviol_pct = -1;         // power-throttle residency % over the window
     double tviol_pct = -1;         // thermal-throttle residency % over the window
     long long samples_taken = 0;

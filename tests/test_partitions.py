@@ -168,3 +168,22 @@ class TestBackendsOnGpu:
         un = {s.labels["counter"] for s in samples
               if s.name == "amd_counter_unavailable"}
         assert info or "compute_partition" in un or "xcp_busy" in un
+
+
+@needs_bin
+class TestExporterObservability:
+    def test_sample_duration_metric(self):
+        with ExporterProcess(mock_devices=2, interval_ms=100) as exp:
+            time.sleep(0.4)
+            samples = scrape(exp)
+        dur = [s for s in samples
+               if s.name == "amd_exporter_sample_duration_ms"]
+        assert len(dur) == 2
+        assert all(0 <= s.value < 1000 for s in dur)
+
+    def test_backend_env_fallback(self, monkeypatch):
+        monkeypatch.setenv("MI355X_EXPORTER_BACKEND", "nonsense")
+        p = subprocess.run([EXPORTER, "--mock", "1"], capture_output=True,
+                           timeout=10)
+        assert p.returncode == 2
+        assert b"bad backend" in p.stderr
