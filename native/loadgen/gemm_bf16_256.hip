@@ -991,6 +991,212 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9w(
     gemm_bf16_tn_256_impl9<1, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
+// ---------------------------------------------------------------------------
+// d18: d9 + 3-deep B rotation in the spare 32 KiB of LDS (160 KiB total).
+//
+// d9's residual wait is its boundary vmcnt(0): the B halves staged at
+// q1/q2 are only ~0.5-1 us old when the drain hits. MI355X has 160 KiB of
+// LDS per CU and the d9 layout uses 128: spending the spare 32 KiB on a
+// THIRD pair of B slots lets B stage TWO tiles ahead (d6's depth) while
+// keeping d9's zero mid-tile barriers:
+//   A: 2 buffers x 2 halves (64 KiB), abuf = kt & 1, staged one ahead at q0
+//   B: 3 buffers x 2 halves (96 KiB), bbuf = kt % 3, staged two ahead at
+//      q1/q2 into (kt+2) % 3 — that slot was consumed at tile kt-1 and its
+//      reads retired before the kt-1 -> kt boundary barrier, so it is dead
+//      with NO mid-tile publish point.
+// Boundary drain becomes vmcnt(4): retires A(kt+1) (staged q0, ~1.5 us
+// old) and B(kt+2) (staged LAST tile, ~3.5 us old) — both long landed —
+// while this tile's B(kt+3) DMAs stay in flight across the barrier.
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl18(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[(4 + 6) * HALF_HW];  // 160 KiB
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;
+    const int wc = w & 3;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    auto frag_off = [&](int row_in_half, int ks) {
+        return swz256(row_in_half * 128 + ks * 64 + ((lane >> 4) * 16));
+    };
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        // A slots: lds[(abuf*2 + h) * HALF_HW], h 0/1
+        // B slots: lds[(4 + bbuf*2 + hb) * HALF_HW], hb 0/1
+        auto stage_a = [&](int kt, int h, int abuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = A + (row0 + h * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(abuf * 2 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+        auto stage_b = [&](int kt, int hb, int bbuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = Bt + (col0 + hb * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(4 + bbuf * 2 + hb) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage_a(0, 0, 0);
+        stage_a(0, 1, 0);
+        stage_b(0, 0, 0);
+        stage_b(0, 1, 0);
+        stage_b(1, 0, 1);
+        stage_b(1, 1, 1);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[2][2];
+        bf16x8 bfrag[4][2];
+
+        int bbuf = 0;       // kt % 3 without a divide
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int abuf = kt & 1;
+            const int bnext2 = bbuf + 2 >= 3 ? bbuf - 1 : bbuf + 2;  // (kt+2)%3
+            const unsigned short* la = &lds[(abuf * 2 + wr) * HALF_HW];
+            const unsigned short* lb =
+                &lds[(4 + bbuf * 2 + (wc >> 1)) * HALF_HW];
+            const int bcol0 = (wc & 1) * 64;
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = q * 2;
+#pragma unroll
+                for (int m = 0; m < 2; ++m) {
+                    const int row = (mbase + m) * 16 + (lane & 15);
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        afrag[m][ks] =
+                            *(const bf16x8*)((const char*)la + frag_off(row, ks));
+                }
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 4; ++n) {
+                        const int col = bcol0 + n * 16 + (lane & 15);
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            bfrag[n][ks] = *(const bf16x8*)((const char*)lb +
+                                                            frag_off(col, ks));
+                    }
+                }
+
+                if (q == 0) {
+                    stage_a(kt + 1, 0, abuf ^ 1);
+                    stage_a(kt + 1, 1, abuf ^ 1);
+                } else if (q == 1) {
+                    stage_b(kt + 2, 0, bnext2);
+                } else if (q == 2) {
+                    stage_b(kt + 2, 1, bnext2);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            acc[mbase + m][n] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag[m][ks], bfrag[n][ks],
+                                    acc[mbase + m][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            // boundary: A(kt+1) + B(kt+2) long landed; B(kt+3) stays in
+            // flight across the barrier (FIFO: 4 newest glds)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            bbuf = bbuf + 1 >= 3 ? 0 : bbuf + 1;
+        }
+
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d18(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl18<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
 // d9 with all four stages issued at q0 — A/B candidate.
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9e(
     const unsigned short* A, const unsigned short* Bt, float* C,
