@@ -83,6 +83,15 @@ def _load():
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double),
         ]
+        lib.lg_gemm_fp8_verify_variant.argtypes = [
+            ctypes.c_int,
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ]
         lib.lg_gemm_fp8_verify.argtypes = [
             ctypes.c_int,
             np.ctypeslib.ndpointer(np.float32, flags="C_CONTIGUOUS"),
@@ -157,7 +166,8 @@ def gemm_bench(m=4096, n=4096, k=4096, warmup=5, iters=50, device=0,
     return ms.value, tf.value
 
 
-def gemm_fp8(a: np.ndarray, bt: np.ndarray, device: int = 0):
+def gemm_fp8(a: np.ndarray, bt: np.ndarray, device: int = 0,
+             raster: int = 1):
     """FP8 (E4M3) MFMA GEMM numerics entry: quantizes the f32 inputs to
     E4M3, computes C = Aq @ Btq^T on the GPU, and returns
     (c, aq, btq) where aq/btq are the dequantized (f32) operands the GPU
@@ -171,7 +181,8 @@ def gemm_fp8(a: np.ndarray, bt: np.ndarray, device: int = 0):
     c = np.empty((m, n), np.float32)
     aq = np.empty_like(a)
     btq = np.empty_like(bt)
-    _check(_load().lg_gemm_fp8_verify(device, a, bt, c, aq, btq, m, n, k))
+    _check(_load().lg_gemm_fp8_verify_variant(device, a, bt, c, aq, btq,
+                                              m, n, k, raster))
     return c, aq, btq
 
 

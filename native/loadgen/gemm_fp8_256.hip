@@ -35,14 +35,17 @@ static __device__ __forceinline__ int swz256b(int byte_off)
     return byte_off ^ (((byte_off >> 9) & 1) << 5);
 }
 
-template <int RASTER = 1>
+// DEEPB=1: 3-deep B rotation in 160 KiB LDS (the bf16 d18 structure —
+// B staged two tiles ahead into a slot dead since the previous boundary
+// barrier; boundary drain vmcnt(4) over long-landed DMAs).
+template <int RASTER = 1, int DEEPB = 0>
 __device__ __forceinline__ void gemm_fp8_tn_256_impl(
     const unsigned char* __restrict__ A,   // [M][K] fp8 E4M3
     const unsigned char* __restrict__ Bt,  // [N][K] fp8 E4M3
     float* __restrict__ C,                 // [M][N] f32
     int M, int N, int K, int tiles_per_cta)
 {
-    __shared__ unsigned char lds[2 * 4 * F8_HALF_BYTES];
+    __shared__ unsigned char lds[(DEEPB ? 10 : 8) * F8_HALF_BYTES];
 
     const int tid = threadIdx.x;
     const int w = tid >> 6;
@@ -99,13 +102,19 @@ __device__ __forceinline__ void gemm_fp8_tn_256_impl(
 #pragma unroll
             for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+        // slot layout: DEEPB=0: [buf][A0 A1 B0 B1] x2 (d9 layout);
+        // DEEPB=1: A [abuf][h] in slots 0-3, B [bbuf][hb] in slots 4-9
+        auto slot = [&](int h, int buf) {
+            return DEEPB ? (h < 2 ? buf * 2 + h : 4 + buf * 2 + (h - 2))
+                         : buf * 4 + h;
+        };
         auto stage = [&](int kt, int h, int buf) {
             if (kt >= kTiles) kt = kTiles - 1;  // tail clamp (benign restage)
             const long k0 = (long)kt * 128 + src_kb;
             const unsigned char* src =
                 (h < 2) ? A + (row0 + h * 128) * (long)K + k0
                         : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
-            unsigned char* dst = &lds[(buf * 4 + h) * F8_HALF_BYTES];
+            unsigned char* dst = &lds[slot(h, buf) * F8_HALF_BYTES];
 #pragma unroll
             for (int it = 0; it < 2; ++it) {
                 const int p = w * 2 + it;
@@ -122,6 +131,10 @@ __device__ __forceinline__ void gemm_fp8_tn_256_impl(
         stage(0, 1, 0);
         stage(0, 2, 0);
         stage(0, 3, 0);
+        if (DEEPB) {
+            stage(1, 2, 1);
+            stage(1, 3, 1);
+        }
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
 
@@ -140,11 +153,14 @@ __device__ __forceinline__ void gemm_fp8_tn_256_impl(
             return r;
         };
 
+        int bbuf = 0;  // kt % 3 without a divide (DEEPB only)
         for (int kt = 0; kt < kTiles; ++kt) {
             const int buf = kt & 1;
-            const unsigned char* la = &lds[(buf * 4 + wr) * F8_HALF_BYTES];
+            const int bnext = DEEPB ? (bbuf + 2 >= 3 ? bbuf - 1 : bbuf + 2)
+                                    : buf ^ 1;  // B dest for kt+2 / kt+1
+            const unsigned char* la = &lds[slot(wr, buf) * F8_HALF_BYTES];
             const unsigned char* lb =
-                &lds[(buf * 4 + 2 + (wc >> 1)) * F8_HALF_BYTES];
+                &lds[slot(2 + (wc >> 1), DEEPB ? bbuf : buf) * F8_HALF_BYTES];
             const int bcol0 = (wc & 1) * 64;
 
 #pragma unroll
@@ -165,9 +181,9 @@ __device__ __forceinline__ void gemm_fp8_tn_256_impl(
                     stage(kt + 1, 0, buf ^ 1);
                     stage(kt + 1, 1, buf ^ 1);
                 } else if (q == 1) {
-                    stage(kt + 1, 2, buf ^ 1);
+                    stage(kt + (DEEPB ? 2 : 1), 2, bnext);
                 } else if (q == 2) {
-                    stage(kt + 1, 3, buf ^ 1);
+                    stage(kt + (DEEPB ? 2 : 1), 3, bnext);
                 }
 
                 __builtin_amdgcn_s_setprio(1);
@@ -183,8 +199,13 @@ __device__ __forceinline__ void gemm_fp8_tn_256_impl(
                                 0, 0, 0, 0x7F, 0, 0x7F);
                 __builtin_amdgcn_s_setprio(0);
             }
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            // boundary: DEEPB leaves this tile's B(kt+2) DMAs in flight
+            if (DEEPB)
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             __builtin_amdgcn_s_barrier();
+            if (DEEPB) bbuf = bbuf + 1 >= 3 ? 0 : bbuf + 1;
         }
 
 #pragma unroll
@@ -216,4 +237,12 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_fp8_tn_256_nr(
     int M, int N, int K, int tiles_per_cta)
 {
     gemm_fp8_tn_256_impl<0>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
+// 3-deep B rotation (160 KiB LDS) — the bf16 d18 structure on fp8.
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_fp8_tn_256_db(
+    const unsigned char* A, const unsigned char* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_fp8_tn_256_impl<1, 1>(A, Bt, C, M, N, K, tiles_per_cta);
 }

@@ -56,6 +56,9 @@ extern "C" __global__ void gemm_bf16_tn_256(const unsigned short*,
 extern "C" __global__ void gemm_fp8_tn_256(const unsigned char*,
                                            const unsigned char*, float*, int,
                                            int, int, int);
+extern "C" __global__ void gemm_fp8_tn_256_db(const unsigned char*,
+                                              const unsigned char*, float*, int,
+                                              int, int, int);
 extern "C" __global__ void gemm_fp8_tn_256_nr(const unsigned char*,
                                               const unsigned char*, float*, int,
                                               int, int, int);
@@ -537,7 +540,10 @@ static int fp8_gemm_launch(const Fp8GemmBufs& g, hipStream_t stream, int raster)
     int n_tiles = (g.m / 256) * (g.n / 256);
     int blocks = n_tiles < 2048 ? n_tiles : 2048;
     int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
-    hipLaunchKernelGGL(raster ? gemm_fp8_tn_256 : gemm_fp8_tn_256_nr,
+    // raster: 1 = product (4x4 super-tile), 0 = linear, 2 = deep-B rotation
+    hipLaunchKernelGGL(raster == 2   ? gemm_fp8_tn_256_db
+                       : raster == 1 ? gemm_fp8_tn_256
+                                     : gemm_fp8_tn_256_nr,
                        dim3(blocks), dim3(512), 0, stream, g.a, g.bt, g.c,
                        g.m, g.n, g.k, tiles_per_cta);
     return 0;
@@ -599,6 +605,38 @@ int lg_gemm_fp8_verify(int device, const float* a_h, const float* bt_h,
     }
     LG_CHECK(hipMemcpy(g.bt, q.data(), q.size(), hipMemcpyHostToDevice));
     fp8_gemm_launch(g, 0, 1);
+    LG_CHECK(hipDeviceSynchronize());
+    LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
+    fp8_gemm_free(g);
+    return 0;
+}
+
+// variant form: raster 0/1/2 selects the kernel (2 = deep-B rotation).
+int lg_gemm_fp8_verify_variant(int device, const float* a_h, const float* bt_h,
+                               float* c_out, float* aq_out, float* btq_out,
+                               int m, int n, int k, int raster)
+{
+    if (m % 256 || n % 256 || k % 128) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "fp8 gemm dims must be multiples of 256/256/128");
+        return -1;
+    }
+    LG_CHECK(hipSetDevice(device));
+    Fp8GemmBufs g;
+    if (fp8_gemm_alloc(g, m, n, k, false)) return -1;
+    std::vector<unsigned char> q((size_t)m * k);
+    for (size_t i = 0; i < q.size(); ++i) {
+        q[i] = f32_to_fp8_e4m3(a_h[i]);
+        if (aq_out) aq_out[i] = fp8_e4m3_to_f32(q[i]);
+    }
+    LG_CHECK(hipMemcpy(g.a, q.data(), q.size(), hipMemcpyHostToDevice));
+    q.resize((size_t)n * k);
+    for (size_t i = 0; i < q.size(); ++i) {
+        q[i] = f32_to_fp8_e4m3(bt_h[i]);
+        if (btq_out) btq_out[i] = fp8_e4m3_to_f32(q[i]);
+    }
+    LG_CHECK(hipMemcpy(g.bt, q.data(), q.size(), hipMemcpyHostToDevice));
+    fp8_gemm_launch(g, 0, raster);
     LG_CHECK(hipDeviceSynchronize());
     LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
     fp8_gemm_free(g);

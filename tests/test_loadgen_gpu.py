@@ -185,9 +185,10 @@ def test_bw_burn_hits_hbm(gpu):
     assert gbps > 3000, f"triad only {gbps:.0f} GB/s"
 
 
+@pytest.mark.parametrize("raster", [1, 2])  # product, deep-B rotation
 @pytest.mark.parametrize("m,n,k", [(256, 256, 128), (256, 256, 256),
                                    (512, 512, 1024), (1024, 1024, 2048)])
-def test_gemm_fp8_numerics(gpu, m, n, k):
+def test_gemm_fp8_numerics(gpu, m, n, k, raster):
     """FP8 (E4M3) MFMA kernel vs an exact fp32 reference over the
     dequantized operands (fp8 products are exact in f32; only the f32
     accumulation rounds, so the tolerance is tight)."""
@@ -197,7 +198,7 @@ def test_gemm_fp8_numerics(gpu, m, n, k):
     bt = rng.uniform(-1, 1, (n, k)).astype(np.float32)
     a[:, 0] += np.arange(m) * 0.01
     bt[:, 0] -= np.arange(n) * 0.01
-    c, aq, btq = lg.gemm_fp8(a, bt)
+    c, aq, btq = lg.gemm_fp8(a, bt, raster=raster)
     # quantization sanity: E4M3 RNE error <= half a step = |x|/32 for
     # normals (3 mantissa bits), 2^-10 floor in the subnormal band
     assert (np.abs(aq - a) <= np.maximum(np.abs(a) / 16 + 1e-6,

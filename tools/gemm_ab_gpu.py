@@ -41,10 +41,10 @@ def main():
         res = {v: [] for v in variants}
         for r in range(rounds):
             for v in variants:
-                if v >= 100:  # 100/101: fp8 kernel raster on/off
-                    ms, tf = loadgen.gemm_fp8_bench(m, n, k, warmup=2,
-                                                    iters=4,
-                                                    raster=(v == 100))
+                if v >= 100:  # 100/101/102: fp8 raster / no-raster / deep-B
+                    ms, tf = loadgen.gemm_fp8_bench(
+                        m, n, k, warmup=2, iters=4,
+                        raster={100: 1, 101: 0, 102: 2}[v])
                 else:
                     ms, tf = loadgen.gemm_bench(m, n, k, warmup=2, iters=4,
                                                 variant=v)
