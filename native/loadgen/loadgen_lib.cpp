@@ -225,6 +225,10 @@ static int gemm_launch(const GemmBufs& g, hipStream_t stream, int variant = 1)
         int n_tiles = (g.m / 256) * (g.n / 256);
         int blocks = n_tiles < 2048 ? n_tiles : 2048;
         int tiles_per_cta = (n_tiles + blocks - 1) / blocks;
+        // product auto-select: at huge grids (>2048 tiles, 16k-class
+        // shapes) the 3-deep-B d18 wins by ~16% (DMA latency under heavy
+        // memory-system load); at <=8192-class shapes d9 ties or wins.
+        if (variant == 2 && n_tiles > 2048) variant = 17;
         hipLaunchKernelGGL(variant == 3 ? gemm_bf16_tn_256_d1 : variant == 4 ? gemm_bf16_tn_256_d4 : variant == 5 ? gemm_bf16_tn_256_d5 : variant == 6 ? gemm_bf16_tn_256_d2 : variant == 7 ? gemm_bf16_tn_256_d7 : variant == 8 ? gemm_bf16_tn_256_d8 : variant == 9 ? gemm_bf16_tn_256_w32 : variant == 10 ? gemm_bf16_tn_256_soft : variant == 11 ? gemm_bf16_tn_256_d9 : variant == 12 ? gemm_bf16_tn_256_d9nr : variant == 14 ? gemm_bf16_tn_256_d6 : variant == 15 ? gemm_bf16_tn_256_d9w : variant == 16 ? gemm_bf16_tn_256_d9e : variant == 17 ? gemm_bf16_tn_256_d18 : gemm_bf16_tn_256, dim3(blocks), dim3(512), 0, stream,
                            g.a, g.bt, g.c, g.m, g.n, g.k, tiles_per_cta);
         return 0;
