@@ -315,14 +315,27 @@ def main():
     load_step_detection_s = None
     detect_threshold = args.load_util / 2.0
     if rank == 0 and has_gpu and load_thread:
+        # stop ONLY rank 0's burn; with world > 1 the deployment average
+        # cannot cross the threshold (7 of 8 pods stay loaded), so the
+        # falling edge is detected on GPU 0's own series instead.
         t_stop = time.monotonic()
         stop_flag.value = 1
         deadline = t_stop + 15.0
         lt = exporter_tick()
+
+        def falling_value():
+            if world == 1:
+                return loop.step().metric_value
+            for s in parse_prometheus_text(exporter.scrape()):
+                if (s.name == "dcgm_gpu_utilization"
+                        and s.labels.get("gpu") == "0"):
+                    return s.value
+            return None
+
         while time.monotonic() < deadline:
             lt = wait_fresh_tick(lt)
-            r = loop.step()
-            if r.metric_value is not None and r.metric_value < detect_threshold:
+            v = falling_value()
+            if v is not None and v < detect_threshold:
                 load_step_detection_s = time.monotonic() - t_stop
                 break
         if load_step_detection_s is not None:
@@ -332,6 +345,10 @@ def main():
             log("load-step NOT detected within 15 s")
 
     # --- stop load + report ----------------------------------------------
+    # non-zero ranks hold their burns until rank 0 finishes the util-err
+    # and detection measurements above (otherwise GPUs 1..N-1 go idle
+    # mid-measurement and the per-GPU util-err evidence is trivial)
+    barrier()
     stop_flag.value = 1
     if load_thread:
         load_thread.join(timeout=10)
