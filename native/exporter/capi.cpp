@@ -106,15 +106,18 @@ int mi355x_render_mock_metrics(int n_devices, const char* attr_key,
                     buflen);
 }
 
-// Benchmark the real rsmi sampling path: n full-device sampling passes.
-// Returns 0 and fills stats (mean/p50/max microseconds per pass + device
-// count), or -1 with the error in errbuf (no GPU / no rocm_smi).
-int mi355x_sample_benchmark(int n, double* mean_us, double* p50_us,
-                            double* max_us, int* n_devices, char* errbuf,
-                            int errlen)
+// Benchmark a real sampling path: n full-device sampling passes.
+// backend_name: "rsmi" / "amdsmi". Returns 0 and fills stats (mean/p50/max
+// microseconds per pass + device count), or -1 with the error in errbuf.
+int mi355x_sample_benchmark_backend(const char* backend_name, int n,
+                                    double* mean_us, double* p50_us,
+                                    double* max_us, int* n_devices,
+                                    char* errbuf, int errlen)
 {
     std::string err;
-    auto backend = mi355x::make_rsmi_backend(&err);
+    auto backend = (backend_name && !std::strcmp(backend_name, "amdsmi"))
+                       ? mi355x::make_amdsmi_backend(&err)
+                       : mi355x::make_rsmi_backend(&err);
     if (!backend) {
         std::snprintf(errbuf, errlen, "%s", err.c_str());
         return -1;
@@ -136,6 +139,14 @@ int mi355x_sample_benchmark(int n, double* mean_us, double* p50_us,
     if (max_us) *max_us = us.back();
     if (n_devices) *n_devices = backend->device_count();
     return 0;
+}
+
+int mi355x_sample_benchmark(int n, double* mean_us, double* p50_us,
+                            double* max_us, int* n_devices, char* errbuf,
+                            int errlen)
+{
+    return mi355x_sample_benchmark_backend("rsmi", n, mean_us, p50_us, max_us,
+                                           n_devices, errbuf, errlen);
 }
 
 } // extern "C"

@@ -1,6 +1,8 @@
-"""GPU test: the rsmi sampling pass must be fast enough to sustain a far
-faster cadence than the reference's 10 s tick (we default the DaemonSet to
-1 s and bench at 100 ms — the sampling pass itself must be milliseconds)."""
+"""GPU test: each native sampling backend's pass must be fast enough to
+sustain a far faster cadence than the reference's 10 s tick (we default
+the DaemonSet to 1 s and bench at 100 ms — the sampling pass itself must
+be milliseconds). Parametrized over both backends (amd-smi is the auto
+default, rocm_smi the fallback)."""
 
 import ctypes
 
@@ -11,9 +13,10 @@ from mi355x_gpu_hpa import NATIVE_BUILD
 pytestmark = pytest.mark.gpu
 
 
-def test_sampling_pass_fast_enough(gpu):
+def _bench(backend: str):
     lib = ctypes.CDLL(str(NATIVE_BUILD / "libmi355x_sampler.so"))
-    lib.mi355x_sample_benchmark.argtypes = [
+    lib.mi355x_sample_benchmark_backend.argtypes = [
+        ctypes.c_char_p,
         ctypes.c_int, ctypes.POINTER(ctypes.c_double),
         ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double),
         ctypes.POINTER(ctypes.c_int), ctypes.c_char_p, ctypes.c_int,
@@ -23,12 +26,18 @@ def test_sampling_pass_fast_enough(gpu):
     mx = ctypes.c_double()
     nd = ctypes.c_int()
     err = ctypes.create_string_buffer(512)
-    rc = lib.mi355x_sample_benchmark(50, ctypes.byref(mean), ctypes.byref(p50),
-                                     ctypes.byref(mx), ctypes.byref(nd), err,
-                                     len(err))
-    assert rc == 0, err.value
-    assert nd.value >= 1
-    print(f"sampling pass over {nd.value} GPU(s): mean {mean.value/1e3:.2f} ms "
-          f"p50 {p50.value/1e3:.2f} ms max {mx.value/1e3:.2f} ms")
+    rc = lib.mi355x_sample_benchmark_backend(
+        backend.encode(), 50, ctypes.byref(mean), ctypes.byref(p50),
+        ctypes.byref(mx), ctypes.byref(nd), err, len(err))
+    assert rc == 0, (backend, err.value)
+    return mean.value, p50.value, mx.value, nd.value
+
+
+@pytest.mark.parametrize("backend", ["rsmi", "amdsmi"])
+def test_sampling_pass_fast_enough(gpu, backend):
+    mean, p50, mx, nd = _bench(backend)
+    assert nd >= 1
+    print(f"{backend} sampling pass over {nd} GPU(s): mean {mean/1e3:.2f} ms "
+          f"p50 {p50/1e3:.2f} ms max {mx/1e3:.2f} ms")
     # must sustain a 100 ms cadence with plenty of margin per GPU
-    assert p50.value / 1e3 < 50 * nd.value, "sampling pass too slow"
+    assert p50 / 1e3 < 50 * nd, f"{backend} sampling pass too slow"
