@@ -1197,6 +1197,223 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d18(
     gemm_bf16_tn_256_impl18<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
+// ---------------------------------------------------------------------------
+// d19: d18 + linearized LDS addressing + software-pipelined A reads.
+//
+// ISA inspection of d9/d18 shows each phase's MFMA block opens with a full
+// s_waitcnt lgkmcnt(0): the phase's fragment reads issue immediately
+// before their consumers, so every phase eats the whole LDS latency. Two
+// changes remove that:
+//  1. The st_16x32 swizzle XOR depends only on bits 0-10 of the byte
+//     offset, i.e. NOT on the fragment index (mf*2048 or nf*2048) — so
+//     swz(x) = frag*2048 + swz(base(lane, ks)). All 24 fragment reads
+//     collapse onto TWO per-lane base offsets with compile-time immediate
+//     deltas (the compiler previously held ~24 hoisted address VGPRs).
+//  2. The freed registers fund a double-buffered afrag: phase q issues
+//     phase q+1's A reads BEFORE its own MFMA segment, so the reads
+//     retire under the matrix math (counted lgkm waits, ~0 exposed
+//     latency for q1-q3).
+// B stays read-at-q0 (its 8 reads overlap q1's prefetched As poorly only
+// at q0; with the deep-B rotation the data has been resident for a full
+// tile, so only LDS latency — not DMA — is exposed there).
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl19(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[(4 + 6) * HALF_HW];  // 160 KiB (d18 layout)
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;
+    const int lane = tid & 63;
+    const int wr = w >> 2;
+    const int wc = w & 3;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    // linearized read addressing: swz(frag_row*128 + ks*64 + lg*16) =
+    // frag*2048 + swz_base[ks] for frag_row = frag*16 + (lane&15)
+    const int swz_base0 = swz256((lane & 15) * 128 + ((lane >> 4) * 16));
+    const int swz_base1 = swz256((lane & 15) * 128 + 64 + ((lane >> 4) * 16));
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        auto stage_a = [&](int kt, int h, int abuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = A + (row0 + h * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(abuf * 2 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+        auto stage_b = [&](int kt, int hb, int bbuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = Bt + (col0 + hb * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(4 + bbuf * 2 + hb) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int p = w * 2 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage_a(0, 0, 0);
+        stage_a(0, 1, 0);
+        stage_b(0, 0, 0);
+        stage_b(0, 1, 0);
+        stage_b(1, 0, 1);
+        stage_b(1, 1, 1);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 af[2][2][2];  // [pipe][m][ks] — double-buffered A fragments
+        bf16x8 bfrag[4][2];
+
+        int bbuf = 0;
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int abuf = kt & 1;
+            const int bnext2 = bbuf + 2 >= 3 ? bbuf - 1 : bbuf + 2;
+            const char* la0 = (const char*)&lds[(abuf * 2 + wr) * HALF_HW] +
+                              swz_base0;
+            const char* la1 = (const char*)&lds[(abuf * 2 + wr) * HALF_HW] +
+                              swz_base1;
+            const char* lb0 =
+                (const char*)&lds[(4 + bbuf * 2 + (wc >> 1)) * HALF_HW] +
+                (wc & 1) * 8192 + swz_base0;
+            const char* lb1 =
+                (const char*)&lds[(4 + bbuf * 2 + (wc >> 1)) * HALF_HW] +
+                (wc & 1) * 8192 + swz_base1;
+
+            // preload phase 0's A fragments (m-frags 0, 1)
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
+                af[0][m][0] = *(const bf16x8*)(la0 + m * 2048);
+                af[0][m][1] = *(const bf16x8*)(la1 + m * 2048);
+            }
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+                bfrag[n][0] = *(const bf16x8*)(lb0 + n * 2048);
+                bfrag[n][1] = *(const bf16x8*)(lb1 + n * 2048);
+            }
+
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                const int mbase = q * 2;
+                // pipeline: issue phase q+1's A reads before q's MFMAs
+                if (q < 3) {
+#pragma unroll
+                    for (int m = 0; m < 2; ++m) {
+                        const int fr = (q + 1) * 2 + m;
+                        af[(q + 1) & 1][m][0] =
+                            *(const bf16x8*)(la0 + fr * 2048);
+                        af[(q + 1) & 1][m][1] =
+                            *(const bf16x8*)(la1 + fr * 2048);
+                    }
+                }
+
+                if (q == 0) {
+                    stage_a(kt + 1, 0, abuf ^ 1);
+                    stage_a(kt + 1, 1, abuf ^ 1);
+                } else if (q == 1) {
+                    stage_b(kt + 2, 0, bnext2);
+                } else if (q == 2) {
+                    stage_b(kt + 2, 1, bnext2);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int m = 0; m < 2; ++m)
+#pragma unroll
+                    for (int n = 0; n < 4; ++n)
+#pragma unroll
+                        for (int ks = 0; ks < 2; ++ks)
+                            acc[mbase + m][n] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    af[q & 1][m][ks], bfrag[n][ks],
+                                    acc[mbase + m][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            bbuf = bbuf + 1 >= 3 ? 0 : bbuf + 1;
+        }
+
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 64 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d19(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl19<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
 // d9 with all four stages issued at q0 — A/B candidate.
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9e(
     const unsigned short* A, const unsigned short* Bt, float* C,
