@@ -143,3 +143,41 @@ def test_readyz_real_backend(gpu):
             f"http://127.0.0.1:{exp.port}/readyz", timeout=2
         ) as r:
             assert r.status == 200
+
+
+def test_hbm_bandwidth_metric_tracks_memory_load(gpu):
+    """Config 5's bandwidth axis on real counters: the streaming-triad
+    burn (~4.8 TB/s bursts) must drive amd_hbm_bandwidth_utilization (UMC
+    activity) far above idle — this is the series the multi-metric HPA
+    rule (deploy/multi-metric/) scales on."""
+    from mi355x_gpu_hpa import loadgen
+
+    stop = ctypes.c_int(0)
+    gbps = ctypes.c_double()
+
+    def burn():
+        loadgen._load().lg_bw_burn(
+            0, ctypes.c_double(100.0), ctypes.c_double(20.0),
+            ctypes.c_double(6.0), ctypes.c_double(100.0),
+            ctypes.byref(stop), ctypes.byref(gbps))
+
+    t = threading.Thread(target=burn, daemon=True)
+    t.start()
+    vals = []
+    try:
+        time.sleep(2.0)
+        with ExporterProcess(interval_ms=250) as exp:
+            for _ in range(8):
+                time.sleep(0.5)
+                for s in parse_prometheus_text(exp.scrape()):
+                    if (s.name == "amd_hbm_bandwidth_utilization"
+                            and s.labels["gpu"] == "0"):
+                        vals.append(s.value)
+    finally:
+        stop.value = 1
+        t.join(timeout=15)
+    assert vals, "amd_hbm_bandwidth_utilization not served"
+    mean = sum(vals) / len(vals)
+    # triad at ~60% of the 8 TB/s peak must push UMC activity high;
+    # exact % depends on firmware accounting — require well above idle
+    assert mean > 30, f"UMC activity only {mean}% under ~4.8 TB/s triad"
