@@ -326,7 +326,11 @@ def main():
         def falling_value():
             if world == 1:
                 return loop.step().metric_value
-            for s in parse_prometheus_text(exporter.scrape()):
+            try:
+                samples = parse_prometheus_text(exporter.scrape())
+            except Exception:  # noqa: BLE001 — transient scrape hiccup
+                return None
+            for s in samples:
                 if (s.name == "dcgm_gpu_utilization"
                         and s.labels.get("gpu") == "0"):
                     return s.value

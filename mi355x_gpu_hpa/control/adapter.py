@@ -167,6 +167,12 @@ class Adapter:
         singular = _PLURAL_TO_SINGULAR.get(resource)
         if singular is None:
             raise AdapterError(f"unknown resource {resource!r}")
+        # object/namespace names are interpolated into the PromQL matcher:
+        # restrict to the RFC-1123 charset k8s enforces (also blocks any
+        # quote/backslash injection into the query)
+        for val in (namespace, object_name):
+            if not re.fullmatch(r"[A-Za-z0-9._-]{1,253}", val or ""):
+                raise AdapterError(f"invalid object name {val!r}")
         discovered = discover(self._samples)
         d = discovered.get(metric)
         if d is None or singular not in d.resources:

@@ -183,3 +183,15 @@ class TestLoopThroughAdapter:
         res = loop.step(now_s=0.0)
         assert res.metric_value is None   # HPA reads <unknown>
         assert res.replicas == 1          # no change
+
+
+class TestInputValidation:
+    def test_query_injection_blocked(self):
+        a = Adapter([recorded_reference_series()])
+        for bad in ('cuda-test"} or up{x="', "a\\b", "", "x" * 300):
+            with pytest.raises(AdapterError, match="invalid object name"):
+                a.get_object_metric("default", "deployments", bad,
+                                    "cuda_test_gpu_avg")
+        with pytest.raises(AdapterError, match="invalid object name"):
+            a.get_object_metric('d"efault', "deployments", "cuda-test",
+                                "cuda_test_gpu_avg")
