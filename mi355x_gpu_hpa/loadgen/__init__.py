@@ -78,6 +78,7 @@ def _load():
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_double, ctypes.c_void_p,
         ]
+        lib.lg_gemm_fp8_burn.argtypes = lib.lg_gemm_burn.argtypes
         lib.lg_gemm_fp8_bench.argtypes = [
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_int, ctypes.c_int, ctypes.c_int,
@@ -198,10 +199,11 @@ def gemm_fp8_bench(m=8192, n=8192, k=8192, warmup=2, iters=10, raster=1,
 
 def gemm_burn(target_util_pct: float, seconds: float, device: int = 0,
               m: int = 4096, n: int = 4096, k: int = 4096,
-              period_ms: float = 100.0):
-    """Duty-cycled GEMM load at a target GPU-busy percentage."""
-    _check(_load().lg_gemm_burn(device, target_util_pct, seconds,
-                                m, n, k, period_ms, None))
+              period_ms: float = 100.0, fp8: bool = False):
+    """Duty-cycled GEMM load at a target GPU-busy percentage (closed-loop
+    on measured GPU-active time); fp8=True burns with the E4M3 kernel."""
+    fn = _load().lg_gemm_fp8_burn if fp8 else _load().lg_gemm_burn
+    _check(fn(device, target_util_pct, seconds, m, n, k, period_ms, None))
 
 
 def bw_burn(target_util_pct: float, seconds: float, device: int = 0,

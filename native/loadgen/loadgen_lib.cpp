@@ -376,6 +376,61 @@ int lg_bw_burn(int device, double target_util_pct, double seconds, double gb,
     return 0;
 }
 
+} // extern "C"
+
+// Shared closed-loop duty engine: runs `launch4()` (4 queued kernel
+// launches + sync, hipEvent-measured) in busy bursts of duty*period and
+// trims the duty fraction by the measured GPU-active fraction per actual
+// period. Used by the bf16 and fp8 burn entries.
+template <typename LaunchFn>
+static int duty_burn_loop(double target_util_pct, double seconds,
+                          double period_ms, volatile int* stop_flag,
+                          LaunchFn&& launch4)
+{
+    hipEvent_t ev0, ev1;
+    LG_CHECK(hipEventCreate(&ev0));
+    LG_CHECK(hipEventCreate(&ev1));
+    double duty = target_util_pct / 100.0;
+    const double duty_lo = duty > 0.25 ? duty - 0.25 : 0.0;
+    const double duty_hi = duty + 0.25 < 1.0 ? duty + 0.25 : 1.0;
+    const double kI = 0.004;
+    double ema = -1;
+    double t_end = now_ms() + seconds * 1e3;
+    while (now_ms() < t_end) {
+        if (stop_flag && *stop_flag) break;
+        double period_start = now_ms();
+        double busy_until = period_start + period_ms * duty;
+        float active_ms = 0;
+        while (now_ms() < busy_until) {
+            LG_CHECK(hipEventRecord(ev0, 0));
+            launch4();
+            LG_CHECK(hipEventRecord(ev1, 0));
+            LG_CHECK(hipDeviceSynchronize());
+            float dt = 0;
+            LG_CHECK(hipEventElapsedTime(&dt, ev0, ev1));
+            active_ms += dt;
+        }
+        if (target_util_pct > 0 && target_util_pct < 100) {
+            double actual_ms = now_ms() - period_start;
+            if (actual_ms < period_ms) actual_ms = period_ms;
+            double active_pct = active_ms / actual_ms * 100.0;
+            ema = ema < 0 ? active_pct : 0.7 * ema + 0.3 * active_pct;
+            duty += kI * (target_util_pct - ema);
+            if (duty < duty_lo) duty = duty_lo;
+            if (duty > duty_hi) duty = duty_hi;
+        }
+        double rest = period_start + period_ms - now_ms();
+        if (rest > 0)
+            std::this_thread::sleep_for(
+                std::chrono::duration<double, std::milli>(rest));
+    }
+    (void)hipEventDestroy(ev0);
+    (void)hipEventDestroy(ev1);
+    return 0;
+}
+
+extern "C" {
+
 // Duty-cycled GEMM burn: aim at `target_util_pct` GPU-busy for `seconds`.
 // Duty cycle over a `period_ms` window: run GEMM launches for duty*period,
 // sleep the rest. Two mechanisms close the gap between wall-clock duty and
@@ -406,56 +461,16 @@ int lg_gemm_burn(int device, double target_util_pct, double seconds,
     LG_CHECK(hipSetDevice(device));
     GemmBufs g;
     if (gemm_alloc(g, m, n, k, true)) return -1;
-    hipEvent_t ev0, ev1;
-    LG_CHECK(hipEventCreate(&ev0));
-    LG_CHECK(hipEventCreate(&ev1));
     // one calibration launch so the first period isn't all compile/warmup
     gemm_launch(g, 0, burn_variant);
     LG_CHECK(hipDeviceSynchronize());
-
-    double duty = target_util_pct / 100.0;
-    // integral trim bounds: the controller may shift duty by at most
-    // +/-25pp from the open-loop setpoint (a foreign load on the same GPU
-    // must not drag it to zero)
-    const double duty_lo = duty > 0.25 ? duty - 0.25 : 0.0;
-    const double duty_hi = duty + 0.25 < 1.0 ? duty + 0.25 : 1.0;
-    const double kI = 0.004;   // duty fraction per %-error per period
-    double ema = -1;           // smoothed measured active% over the period
-
-    double t_end = now_ms() + seconds * 1e3;
-    while (now_ms() < t_end) {
-        if (stop_flag && *stop_flag) break;
-        double period_start = now_ms();
-        double busy_until = period_start + period_ms * duty;
-        float active_ms = 0;
-        while (now_ms() < busy_until) {
-            LG_CHECK(hipEventRecord(ev0, 0));
-            for (int b = 0; b < 4; ++b) gemm_launch(g, 0, burn_variant);
-            LG_CHECK(hipEventRecord(ev1, 0));
-            LG_CHECK(hipDeviceSynchronize());
-            float dt = 0;
-            LG_CHECK(hipEventElapsedTime(&dt, ev0, ev1));
-            active_ms += dt;
-        }
-        if (target_util_pct > 0 && target_util_pct < 100) {
-            // denominator: the period's ACTUAL length (the last batch can
-            // overrun busy_until; the nominal period would overstate busy)
-            double actual_ms = now_ms() - period_start;
-            if (actual_ms < period_ms) actual_ms = period_ms;
-            double active_pct = active_ms / actual_ms * 100.0;
-            ema = ema < 0 ? active_pct : 0.7 * ema + 0.3 * active_pct;
-            duty += kI * (target_util_pct - ema);
-            if (duty < duty_lo) duty = duty_lo;
-            if (duty > duty_hi) duty = duty_hi;
-        }
-        double rest = period_start + period_ms - now_ms();
-        if (rest > 0)
-            std::this_thread::sleep_for(std::chrono::duration<double, std::milli>(rest));
-    }
-    (void)hipEventDestroy(ev0);
-    (void)hipEventDestroy(ev1);
+    int rc = duty_burn_loop(target_util_pct, seconds, period_ms, stop_flag,
+                            [&] {
+                                for (int b = 0; b < 4; ++b)
+                                    gemm_launch(g, 0, burn_variant);
+                            });
     gemm_free(g);
-    return 0;
+    return rc;
 }
 
 } // extern "C"
@@ -646,6 +661,37 @@ int lg_gemm_fp8_verify_variant(int device, const float* a_h, const float* bt_h,
     LG_CHECK(hipMemcpy(c_out, g.c, (size_t)m * n * 4, hipMemcpyDeviceToHost));
     fp8_gemm_free(g);
     return 0;
+}
+
+// fp8 (E4M3) variant of the burn: same closed-loop duty engine over the
+// fp8 MFMA kernel — the 2x-peak datatype's power/clock profile as a load.
+int lg_gemm_fp8_burn(int device, double target_util_pct, double seconds,
+                     int m, int n, int k, double period_ms,
+                     volatile int* stop_flag)
+{
+    if (m <= 0) m = 4096;
+    if (n <= 0) n = 4096;
+    if (k <= 0) k = 4096;
+    if (period_ms <= 0) period_ms = 100.0;
+    if (m % 256 || n % 256 || k % 128) {
+        std::snprintf(g_last_error, sizeof(g_last_error),
+                      "fp8 gemm dims must be multiples of 256/256/128");
+        return -1;
+    }
+    if (target_util_pct < 0) target_util_pct = 0;
+    if (target_util_pct > 100) target_util_pct = 100;
+    LG_CHECK(hipSetDevice(device));
+    Fp8GemmBufs g;
+    if (fp8_gemm_alloc(g, m, n, k, true)) return -1;
+    fp8_gemm_launch(g, 0, 1);
+    LG_CHECK(hipDeviceSynchronize());
+    int rc = duty_burn_loop(target_util_pct, seconds, period_ms, stop_flag,
+                            [&] {
+                                for (int b = 0; b < 4; ++b)
+                                    fp8_gemm_launch(g, 0, 1);
+                            });
+    fp8_gemm_free(g);
+    return rc;
 }
 
 } // extern "C"

@@ -31,6 +31,9 @@ int lg_gemm_fp8_bench(int device, int m, int n, int k, int warmup, int iters,
                       int raster, double* ms_out, double* tflops_out);
 int lg_gemm_burn(int device, double target_util_pct, double seconds,
                  int m, int n, int k, double period_ms, volatile int* stop_flag);
+int lg_gemm_fp8_burn(int device, double target_util_pct, double seconds,
+                     int m, int n, int k, double period_ms,
+                     volatile int* stop_flag);
 int lg_bw_burn(int device, double target_util_pct, double seconds, double gb,
                double period_ms, volatile int* stop_flag, double* gbps_out);
 }
@@ -119,7 +122,14 @@ int main(int argc, char** argv)
         int n = (int)argd(argc, argv, "--n-dim", 4096);
         int k = (int)argd(argc, argv, "--k", 4096);
         double period = argd(argc, argv, "--period-ms", 100.0);
-        if (lg_gemm_burn(device, util, seconds, m, n, k, period, nullptr)) {
+        bool fp8 = false;
+        for (int i = 1; i < argc; ++i)
+            if (!std::strcmp(argv[i], "--fp8")) fp8 = true;
+        int rc = fp8 ? lg_gemm_fp8_burn(device, util, seconds, m, n, k,
+                                        period, nullptr)
+                     : lg_gemm_burn(device, util, seconds, m, n, k, period,
+                                    nullptr);
+        if (rc) {
             std::fprintf(stderr, "error: %s\n", lg_last_error());
             return 2;
         }
