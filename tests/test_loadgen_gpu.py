@@ -218,3 +218,40 @@ def test_gemm_fp8_bench_sane(gpu):
     ms, tf = lg.gemm_fp8_bench(m=2048, n=2048, k=2048, warmup=2, iters=10)
     assert ms > 0
     assert tf > 200, f"fp8 GEMM at {tf:.0f} TF/s — MFMA fp8 path not engaged?"
+
+
+def test_gemm_fp8_burn_tracks_target(gpu):
+    """The fp8 burn shares the closed-loop duty engine: 50% target must
+    land within the same ±10pp band as the bf16 burn."""
+    import ctypes
+    import threading
+    import time
+
+    from mi355x_gpu_hpa import loadgen
+    from mi355x_gpu_hpa.control import parse_prometheus_text
+    from mi355x_gpu_hpa.exporter import ExporterProcess
+
+    stop = ctypes.c_int(0)
+
+    def burn():
+        loadgen._load().lg_gemm_fp8_burn(
+            0, ctypes.c_double(50.0), ctypes.c_double(25.0),
+            4096, 4096, 4096, ctypes.c_double(100.0), ctypes.byref(stop))
+
+    t = threading.Thread(target=burn, daemon=True)
+    t.start()
+    vals = []
+    try:
+        time.sleep(3.0)
+        with ExporterProcess(interval_ms=250) as exp:
+            for _ in range(8):
+                time.sleep(0.5)
+                for s in parse_prometheus_text(exp.scrape()):
+                    if (s.name == "dcgm_gpu_utilization"
+                            and s.labels["gpu"] == "0"):
+                        vals.append(s.value)
+    finally:
+        stop.value = 1
+        t.join(timeout=15)
+    mean = sum(vals) / len(vals)
+    assert 40 <= mean <= 60, f"fp8 burn mean busy {mean}% for 50% target"
