@@ -1414,6 +1414,182 @@ extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d19(
     gemm_bf16_tn_256_impl19<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
+// ---------------------------------------------------------------------------
+// d20: 4 waves x 128x128 wave tiles — the hipBLASLt/Tensile design point.
+//
+// Kernel-trace of torch.matmul at 8192³ shows the library's winner is
+// MT256x256x64_MI16x16 — OUR macro-tile and MFMA shape — at ~0.80 ms vs
+// our 1.10 (gpurun_out/prof_blaslt). Its structural difference: ONE wave
+// per SIMD with the full 512-register budget (256 arch VGPRs + 256
+// accumulation AGPRs), i.e. 4 waves of 128x128 output each. Effects:
+//   * per-FLOP LDS read traffic HALVES (each A row read by 2 wave-cols
+//     and each B col by 2 wave-rows instead of 4/2): 32 KB per wave per
+//     K-tile x 4 waves = 128 KB vs the 8-wave schedules' 192 KB;
+//   * the MFMA pipe is fed by a single wave with 64 independent acc
+//     chains — no co-resident-wave arbitration, issue slack for reads;
+//   * barriers synchronize 4 waves instead of 8 (less skew).
+// acc[8][8] f32x4 = 256 registers lands in AGPRs (launch_bounds(256,1)
+// gives the 512-reg budget); B frags all held from q0 (64 VGPRs), A
+// fragments read per phase with the d19 linearized addressing. Staging,
+// liveness and the single boundary barrier are d9's, with 4 glds per
+// wave per half (16 KiB / 4 waves).
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl20(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[2 * 4 * HALF_HW];
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;       // 0..3
+    const int lane = tid & 63;
+    const int wr = w >> 1;        // 0..1: A half (rows wr*128..+128)
+    const int wc = w & 1;         // 0..1: B half (cols wc*128..+128)
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    const int swz_base0 = swz256((lane & 15) * 128 + ((lane >> 4) * 16));
+    const int swz_base1 = swz256((lane & 15) * 128 + 64 + ((lane >> 4) * 16));
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][8];  // 256 registers -> AGPR file
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 8; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        // 4 glds per wave per half (pieces w*4 .. w*4+3)
+        auto stage = [&](int kt, int h, int buf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src =
+                (h < 2) ? A + (row0 + h * 128) * (long)K + k0
+                        : Bt + (col0 + (h - 2) * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(buf * 4 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 4; ++it) {
+                const int p = w * 4 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage(0, 0, 0);
+        stage(0, 1, 0);
+        stage(0, 2, 0);
+        stage(0, 3, 0);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 afrag[2];
+        bf16x8 bfrag[8][2];  // all 8 B fragments held from q0 (64 VGPRs)
+
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int buf = kt & 1;
+            const char* la0 = (const char*)&lds[(buf * 4 + wr) * HALF_HW] +
+                              swz_base0;
+            const char* la1 = (const char*)&lds[(buf * 4 + wr) * HALF_HW] +
+                              swz_base1;
+            const char* lb0 = (const char*)&lds[(buf * 4 + 2 + wc) * HALF_HW] +
+                              swz_base0;
+            const char* lb1 = (const char*)&lds[(buf * 4 + 2 + wc) * HALF_HW] +
+                              swz_base1;
+
+#pragma unroll
+            for (int q = 0; q < 8; ++q) {
+                afrag[0] = *(const bf16x8*)(la0 + q * 2048);
+                afrag[1] = *(const bf16x8*)(la1 + q * 2048);
+                if (q == 0) {
+#pragma unroll
+                    for (int n = 0; n < 8; ++n) {
+                        bfrag[n][0] = *(const bf16x8*)(lb0 + n * 2048);
+                        bfrag[n][1] = *(const bf16x8*)(lb1 + n * 2048);
+                    }
+                }
+
+                if (q == 0) {
+                    stage(kt + 1, 0, buf ^ 1);
+                } else if (q == 1) {
+                    stage(kt + 1, 1, buf ^ 1);
+                } else if (q == 2) {
+                    stage(kt + 1, 2, buf ^ 1);
+                } else if (q == 3) {
+                    stage(kt + 1, 3, buf ^ 1);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int n = 0; n < 8; ++n)
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        acc[q][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            afrag[ks], bfrag[n][ks], acc[q][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 128 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256, 1) gemm_bf16_tn_256_d20(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl20<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
 // d9 with all four stages issued at q0 — A/B candidate.
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9e(
     const unsigned short* A, const unsigned short* Bt, float* C,

@@ -7,7 +7,7 @@ DVFS drift hits all variants equally; prints per-run TF/s and the max.
 Variant map: 2=product d6+raster, 7=d7 5-barrier, 8=d6 no-raster,
 10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster, 13=d14
 16-wave 4-waves/SIMD, 14=d6 round-1 product (now ablation),
-17=d18 3-deep-B rotation (160 KiB LDS), 18=d19 pipelined-reads, 100/101=fp8 E4M3 kernel with/without raster.
+17=d18 3-deep-B rotation (160 KiB LDS), 18=d19 pipelined-reads, 19=d20 4-wave AGPR-acc, 100/101=fp8 E4M3 kernel with/without raster.
 """
 
 import json
