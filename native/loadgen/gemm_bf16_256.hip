@@ -1590,6 +1590,204 @@ extern "C" __global__ void __launch_bounds__(256, 1) gemm_bf16_tn_256_d20(
     gemm_bf16_tn_256_impl20<1>(A, Bt, C, M, N, K, tiles_per_cta);
 }
 
+// ---------------------------------------------------------------------------
+// d21: d20 (4 waves, 128x128 tiles, AGPR acc) + the latency-hiding a
+// single wave per SIMD actually needs:
+//   * 3-deep B rotation in 160 KiB LDS (d18): B(kt+1) is PUBLISHED a full
+//     tile early, so the next tile's 16 B-fragment reads prefetch at q7
+//     of the current tile into a double-buffered bfrag set (128 VGPRs);
+//   * A fragments double-buffered and prefetched one phase ahead (d19);
+//     only q0's A reads after the boundary barrier expose LDS latency.
+// With no co-resident wave to cover stalls (d20's regression), every
+// latency must hide inside this wave's own MFMA stream — this is what
+// the Tensile MT256x256x64 kernels do that the plain d20 port did not.
+// ---------------------------------------------------------------------------
+template <int RASTER = 1>
+__device__ __forceinline__ void gemm_bf16_tn_256_impl21(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K, int tiles_per_cta)
+{
+    __shared__ unsigned short lds[(4 + 6) * HALF_HW];  // A 2x2, B 3x2 slots
+
+    const int tid = threadIdx.x;
+    const int w = tid >> 6;       // 0..3
+    const int lane = tid & 63;
+    const int wr = w >> 1;
+    const int wc = w & 1;
+
+    const int n_tiles_n = N / 256;
+    const int n_tiles_m = M / 256;
+    const int n_tiles = n_tiles_m * n_tiles_n;
+    const int kTiles = K / 64;
+
+    const int nwg = gridDim.x;
+    int wgid = blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = wgid & 7, pos = wgid >> 3;
+        wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    }
+
+    const int in_piece = swz256(lane * 16) & 1023;
+    const int src_row = in_piece >> 7;
+    const int src_kk = (in_piece & 127) >> 1;
+
+    const int swz_base0 = swz256((lane & 15) * 128 + ((lane >> 4) * 16));
+    const int swz_base1 = swz256((lane & 15) * 128 + 64 + ((lane >> 4) * 16));
+
+    const bool super4 = RASTER && (n_tiles_n % 4 == 0) && (n_tiles_m % 4 == 0);
+
+    for (int t = 0; t < tiles_per_cta; ++t) {
+        const int tile = wgid + t * nwg;
+        if (tile >= n_tiles) return;
+        int tm, tn;
+        if (super4) {
+            const int sb = tile >> 4, wi = tile & 15;
+            const int sbn = n_tiles_n >> 2;
+            tm = (sb / sbn) * 4 + (wi >> 2);
+            tn = (sb % sbn) * 4 + (wi & 3);
+        } else {
+            tm = tile / n_tiles_n;
+            tn = tile % n_tiles_n;
+        }
+        const long row0 = (long)tm * 256;
+        const long col0 = (long)tn * 256;
+
+        f32x4 acc[8][8];  // 256 registers -> AGPR file
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 8; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+        auto stage_a = [&](int kt, int h, int abuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = A + (row0 + h * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(abuf * 2 + h) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 4; ++it) {
+                const int p = w * 4 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+        auto stage_b = [&](int kt, int hb, int bbuf) {
+            if (kt >= kTiles) kt = kTiles - 1;
+            const long k0 = (long)kt * 64 + src_kk;
+            const unsigned short* src = Bt + (col0 + hb * 128) * (long)K + k0;
+            unsigned short* dst = &lds[(4 + bbuf * 2 + hb) * HALF_HW];
+#pragma unroll
+            for (int it = 0; it < 4; ++it) {
+                const int p = w * 4 + it;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        (src + (long)(p * 8 + src_row) * K),
+                    (__attribute__((address_space(3))) unsigned int*)
+                        (dst + p * 512),
+                    16, 0, 0);
+            }
+        };
+
+        stage_a(0, 0, 0);
+        stage_a(0, 1, 0);
+        stage_b(0, 0, 0);
+        stage_b(0, 1, 0);
+        stage_b(1, 0, 1);
+        stage_b(1, 1, 1);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        bf16x8 af[2][2];      // [pipe][ks]
+        bf16x8 bf[8][2];      // B fragments, read once per tile at q0
+                              // (double-buffering spilled: MFMA operands
+                              // must be arch VGPRs and 128 extra regs do
+                              // not fit beside af + addressing)
+
+        int bbuf = 0;
+        for (int kt = 0; kt < kTiles; ++kt) {
+            const int abuf = kt & 1;
+            const int bnext2 = bbuf + 2 >= 3 ? bbuf - 1 : bbuf + 2;
+            const int bbuf1 = bbuf + 1 >= 3 ? 0 : bbuf + 1;  // (kt+1)%3
+            const char* la0 = (const char*)&lds[(abuf * 2 + wr) * HALF_HW] +
+                              swz_base0;
+            const char* la1 = (const char*)&lds[(abuf * 2 + wr) * HALF_HW] +
+                              swz_base1;
+            const char* lb0 =
+                (const char*)&lds[(4 + bbuf * 2 + wc) * HALF_HW] + swz_base0;
+            const char* lb1 =
+                (const char*)&lds[(4 + bbuf * 2 + wc) * HALF_HW] + swz_base1;
+
+            // q0's fragments (the reads the barrier exposes; B is resident
+            // since a tile ago — deep-B — so this is LDS latency only)
+            af[0][0] = *(const bf16x8*)(la0);
+            af[0][1] = *(const bf16x8*)(la1);
+#pragma unroll
+            for (int n = 0; n < 8; ++n) {
+                bf[n][0] = *(const bf16x8*)(lb0 + n * 2048);
+                bf[n][1] = *(const bf16x8*)(lb1 + n * 2048);
+            }
+
+#pragma unroll
+            for (int q = 0; q < 8; ++q) {
+                // prefetch next phase's A under this phase's MFMAs
+                if (q < 7) {
+                    af[(q + 1) & 1][0] =
+                        *(const bf16x8*)(la0 + (q + 1) * 2048);
+                    af[(q + 1) & 1][1] =
+                        *(const bf16x8*)(la1 + (q + 1) * 2048);
+                }
+
+                if (q == 0) {
+                    stage_a(kt + 1, 0, abuf ^ 1);
+                } else if (q == 1) {
+                    stage_a(kt + 1, 1, abuf ^ 1);
+                } else if (q == 2) {
+                    stage_b(kt + 2, 0, bnext2);
+                } else if (q == 3) {
+                    stage_b(kt + 2, 1, bnext2);
+                }
+
+                __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                for (int n = 0; n < 8; ++n)
+#pragma unroll
+                    for (int ks = 0; ks < 2; ++ks)
+                        acc[q][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af[q & 1][ks], bf[n][ks], acc[q][n], 0, 0, 0);
+                __builtin_amdgcn_s_setprio(0);
+            }
+            asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+            bbuf = bbuf1;
+        }
+
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const long row = row0 + wr * 128 + i * 16 + (lane >> 4) * 4 + r;
+                    const long col = col0 + wc * 128 + j * 16 + (lane & 15);
+                    C[row * (long)N + col] = acc[i][j][r];
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" __global__ void __launch_bounds__(256, 1) gemm_bf16_tn_256_d21(
+    const unsigned short* A, const unsigned short* Bt, float* C,
+    int M, int N, int K, int tiles_per_cta)
+{
+    gemm_bf16_tn_256_impl21<1>(A, Bt, C, M, N, K, tiles_per_cta);
+}
+
 // d9 with all four stages issued at q0 — A/B candidate.
 extern "C" __global__ void __launch_bounds__(512, 2) gemm_bf16_tn_256_d9e(
     const unsigned short* A, const unsigned short* Bt, float* C,

@@ -35,6 +35,7 @@ extern "C" __global__ void gemm_bf16_tn(const unsigned short*, const unsigned sh
 extern "C" __global__ void gemm_bf16_tn_linear(const unsigned short*,
                                                const unsigned short*, float*, int,
                                                int, int, int);
+extern "C" __global__ void gemm_bf16_tn_256_d21(const unsigned short*, const unsigned short*, float*, int, int, int, int);
 extern "C" __global__ void gemm_bf16_tn_256_d20(const unsigned short*, const unsigned short*, float*, int, int, int, int);
 extern "C" __global__ void gemm_bf16_tn_256_d19(const unsigned short*, const unsigned short*, float*, int, int, int, int);
 extern "C" __global__ void gemm_bf16_tn_256_d18(const unsigned short*, const unsigned short*, float*, int, int, int, int);
@@ -231,10 +232,11 @@ static int gemm_launch(const GemmBufs& g, hipStream_t stream, int variant = 1)
         // shapes) the 3-deep-B d18 wins by ~16% (DMA latency under heavy
         // memory-system load); at <=8192-class shapes d9 ties or wins.
         if (variant == 2 && n_tiles > 2048) variant = 17;
-        if (variant == 19) {  // d20: 256-thread CTA, 1 wave/SIMD, AGPR acc
-            hipLaunchKernelGGL(gemm_bf16_tn_256_d20, dim3(blocks), dim3(256),
-                               0, stream, g.a, g.bt, g.c, g.m, g.n, g.k,
-                               tiles_per_cta);
+        if (variant == 19 || variant == 20) {  // d20/d21: 256-thread CTA,
+            hipLaunchKernelGGL(                    // 1 wave/SIMD, AGPR acc
+                variant == 19 ? gemm_bf16_tn_256_d20 : gemm_bf16_tn_256_d21,
+                dim3(blocks), dim3(256), 0, stream, g.a, g.bt, g.c, g.m,
+                g.n, g.k, tiles_per_cta);
             return 0;
         }
         hipLaunchKernelGGL(variant == 3 ? gemm_bf16_tn_256_d1 : variant == 4 ? gemm_bf16_tn_256_d4 : variant == 5 ? gemm_bf16_tn_256_d5 : variant == 6 ? gemm_bf16_tn_256_d2 : variant == 7 ? gemm_bf16_tn_256_d7 : variant == 8 ? gemm_bf16_tn_256_d8 : variant == 9 ? gemm_bf16_tn_256_w32 : variant == 10 ? gemm_bf16_tn_256_soft : variant == 11 ? gemm_bf16_tn_256_d9 : variant == 12 ? gemm_bf16_tn_256_d9nr : variant == 14 ? gemm_bf16_tn_256_d6 : variant == 15 ? gemm_bf16_tn_256_d9w : variant == 16 ? gemm_bf16_tn_256_d9e : variant == 17 ? gemm_bf16_tn_256_d18 : variant == 18 ? gemm_bf16_tn_256_d19 : gemm_bf16_tn_256, dim3(blocks), dim3(512), 0, stream,

@@ -86,7 +86,7 @@ def test_gemm_bf16_256_numerics(gpu, m, n, k):
     np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2 * np.sqrt(k) / 8)
 
 
-@pytest.mark.parametrize("variant", [11, 15, 17, 18, 19])  # d9/d9w/d18/d19/d20
+@pytest.mark.parametrize("variant", [11, 15, 17, 18, 19, 20])  # d9..d21
 @pytest.mark.parametrize("m,n,k", [(256, 256, 64), (256, 256, 128),
                                    (512, 256, 256), (512, 512, 1024),
                                    (1024, 1024, 2048)])
