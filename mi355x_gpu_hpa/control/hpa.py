@@ -39,6 +39,11 @@ class HpaSpec:
     # (v2beta1 / reference behavior).
     scale_up_pods: int = 0
     scale_up_period_s: float = 15.0
+    # v2 behavior.scaleDown Pods policy: at most `scale_down_pods` removed
+    # per `scale_down_period_s` (applied AFTER stabilization, upstream
+    # ordering). 0 = unlimited.
+    scale_down_pods: int = 0
+    scale_down_period_s: float = 15.0
 
 
 @dataclass
@@ -48,6 +53,8 @@ class HpaState:
     recommendations: List[Tuple[float, int]] = field(default_factory=list)
     # (period_start_s, replicas_at_period_start) for the scaleUp policy
     scaleup_window: Tuple[Optional[float], Optional[int]] = (None, None)
+    # same for the scaleDown policy
+    scaledown_window: Tuple[Optional[float], Optional[int]] = (None, None)
 
 
 def desired_replicas(
@@ -148,6 +155,13 @@ def _stabilize(spec: HpaSpec, state: HpaState, desired: int, now_s: float) -> in
             base_t, base_r = now_s, state.current_replicas
         new = min(new, base_r + spec.scale_up_pods)
         state.scaleup_window = (base_t, base_r)
+    if spec.scale_down_pods > 0 and new < state.current_replicas:
+        # v2 scaleDown Pods policy, symmetric to scaleUp
+        base_t, base_r = state.scaledown_window
+        if base_t is None or now_s - base_t >= spec.scale_down_period_s:
+            base_t, base_r = now_s, state.current_replicas
+        new = max(new, base_r - spec.scale_down_pods)
+        state.scaledown_window = (base_t, base_r)
     new = max(spec.min_replicas, min(spec.max_replicas, new))
     state.current_replicas = new
     return new

@@ -141,3 +141,28 @@ class TestReconcile:
             seen.append(st.current_replicas)
         assert st.current_replicas == 8
         assert seen[0] == 1 and sorted(seen) == seen  # monotone ramp
+
+
+class TestScaleDownPolicy:
+    def test_scaledown_pods_ramp(self):
+        """v2 behavior.scaleDown Pods policy: at most N removed per
+        period, applied after stabilization (upstream ordering)."""
+        s = spec(max_replicas=8, downscale_stabilization_s=0.0,
+                 scale_down_pods=1, scale_down_period_s=15.0)
+        st = HpaState(current_replicas=8)
+        counts = [reconcile(s, st, 1.0, now_s=i * 15.0) for i in range(8)]
+        assert counts == [7, 6, 5, 4, 3, 2, 1, 1]
+
+    def test_scaledown_within_period(self):
+        s = spec(max_replicas=8, downscale_stabilization_s=0.0,
+                 scale_down_pods=2, scale_down_period_s=15.0)
+        st = HpaState(current_replicas=8)
+        assert reconcile(s, st, 1.0, now_s=0.0) == 6
+        assert reconcile(s, st, 1.0, now_s=5.0) == 6   # same period
+        assert reconcile(s, st, 1.0, now_s=15.0) == 4
+
+    def test_scaledown_policy_inactive_without_flag(self):
+        s = spec(max_replicas=8, downscale_stabilization_s=0.0)
+        st = HpaState(current_replicas=8)
+        # unlimited drop straight to desired = ceil(8 * 1/5) = 2
+        assert reconcile(s, st, 1.0, now_s=0.0) == 2

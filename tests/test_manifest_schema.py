@@ -102,3 +102,28 @@ class TestValidatorCatchesRealMistakes:
     def test_unpinned_kind_rejected(self):
         errs = validate_manifest({"kind": "FancyNewThing"})
         assert errs and "no pinned schema" in str(errs[0])
+
+
+class TestReferenceManifestsAgainstModernSchema:
+    """The validator applied to the REFERENCE's own manifests: its HPA uses
+    autoscaling/v2beta1 (reference cuda-test-hpa.yaml:1), which modern
+    clusters (>=1.26) reject — exactly the apply-time failure this stack's
+    rewritten manifest avoids. The validator must reproduce that verdict."""
+
+    REF = Path("/root/reference")
+
+    def test_reference_hpa_rejected_for_v2beta1(self):
+        if not (self.REF / "cuda-test-hpa.yaml").exists():
+            pytest.skip("reference tree not present")
+        doc = yaml.safe_load((self.REF / "cuda-test-hpa.yaml").read_text())
+        errors = validate_manifest(doc)
+        assert any("autoscaling/v2beta1" in str(e) for e in errors), errors
+
+    def test_reference_deployment_validates(self):
+        # the workload manifest itself is schema-clean (only the GPU
+        # resource name differs from ours)
+        if not (self.REF / "cuda-test-deployment.yaml").exists():
+            pytest.skip("reference tree not present")
+        doc = yaml.safe_load(
+            (self.REF / "cuda-test-deployment.yaml").read_text())
+        assert validate_manifest(doc) == []
