@@ -125,3 +125,51 @@ def test_malformed_rule_does_not_crash_loop():
     assert "bad_rule" in loop._rule_errors
     # second step: still alive, error reported once
     loop.step()
+
+
+@needs_bin
+def test_two_node_cluster_rule_semantics(tmp_path):
+    """Two exporter daemons = two NODES of the DaemonSet (reference scrape
+    config discovers every endpoint, kube-prometheus-stack-values.yaml:8-12);
+    the rule must average per-pod utilization ACROSS nodes and the node
+    relabel must keep the series distinct before the join."""
+    b1 = tmp_path / "b1"
+    b2 = tmp_path / "b2"
+    b1.write_text("60\n")
+    b2.write_text("20\n")
+    with ExporterProcess(mock_devices=1, interval_ms=50,
+                         mock_busy_file=str(b1)) as e1, \
+         ExporterProcess(mock_devices=1, interval_ms=50,
+                         mock_busy_file=str(b2)) as e2:
+        scraper = Scraper([
+            ScrapeTarget(e1.url, node="node-a",
+                         extra_labels={"pod": "cuda-test-a",
+                                       "namespace": "default"}),
+            ScrapeTarget(e2.url, node="node-b",
+                         extra_labels={"pod": "cuda-test-b",
+                                       "namespace": "default"}),
+        ])
+        loop = ControlLoop(
+            scraper,
+            hpa_spec=HpaSpec(min_replicas=1, max_replicas=8,
+                             target_value=5.0),
+            extra_samples=lambda: synth_pod_labels(
+                ["cuda-test-a", "cuda-test-b"]),
+            use_adapter=True,
+        )
+        time.sleep(0.15)
+        r = loop.step()
+        # avg over the two pods on two nodes: (60 + 20) / 2
+        assert r.metric_value == 40.0
+        assert r.replicas == 8
+        # the raw series keep distinct node labels pre-aggregation
+        nodes = {s.labels.get("node") for s in scraper.scrape_once()
+                 if s.name == "dcgm_gpu_utilization"}
+        assert nodes == {"node-a", "node-b"}
+
+        # one node's exporter dies: its last samples keep serving (stale
+        # handling) and the loop keeps a value rather than going unknown
+        e2.terminate()
+        time.sleep(0.15)
+        r2 = loop.step()
+        assert r2.metric_value is not None
