@@ -13,7 +13,8 @@ Supported grammar:
   * instant vector selectors with label matchers (=, !=, =~, !~)
   * aggregations: sum / avg / max / min / count, with optional ``by (...)``
     (prefix or suffix position)
-  * binary operators: ``*``, ``+``, ``-``, ``/`` with optional
+  * binary operators: ``*``, ``+``, ``-``, ``/`` with * / binding
+    tighter than + - (Prometheus precedence), and optional
     ``on(...)``/``ignoring(...)`` and ``group_left(...)`` modifiers
   * scalar literals (e.g. ``expr * 100``)
   * parentheses
@@ -120,6 +121,7 @@ class Agg:
 class BinOp:
     op: str
     on: Optional[List[str]]
+    ignoring: Optional[List[str]]
     group_left: Optional[List[str]]
     lhs: "Expr"
     rhs: "Expr"
@@ -171,9 +173,9 @@ class _Parser:
             if t is None or t[1] not in ("+", "-"):
                 return lhs
             op = self.next()[1]
-            on, group_left = self.parse_match_modifiers()
+            on, ignoring, group_left = self.parse_match_modifiers()
             rhs = self.parse_multiplicative()
-            lhs = BinOp(op, on, group_left, lhs, rhs)
+            lhs = BinOp(op, on, ignoring, group_left, lhs, rhs)
 
     def parse_multiplicative(self) -> Expr:
         lhs = self.parse_primary()
@@ -182,12 +184,12 @@ class _Parser:
             if t is None or t[1] not in ("*", "/"):
                 return lhs
             op = self.next()[1]
-            on, group_left = self.parse_match_modifiers()
+            on, ignoring, group_left = self.parse_match_modifiers()
             rhs = self.parse_primary()
-            lhs = BinOp(op, on, group_left, lhs, rhs)
+            lhs = BinOp(op, on, ignoring, group_left, lhs, rhs)
 
     def parse_match_modifiers(self):
-        on = group_left = None
+        on = ignoring = group_left = None
         t = self.peek()
         if t and t[0] == "id" and t[1] in ("on", "ignoring"):
             kind = self.next()[1]
@@ -195,7 +197,7 @@ class _Parser:
             if kind == "on":
                 on = names
             else:
-                raise PromQLError("ignoring() not supported; use on()")
+                ignoring = names
         t = self.peek()
         if t and t[0] == "id" and t[1] in ("group_left", "group_right"):
             kind = self.next()[1]
@@ -204,7 +206,7 @@ class _Parser:
             group_left = []
             if self.peek() and self.peek()[1] == "(":
                 group_left = self.parse_name_list()
-        return on, group_left
+        return on, ignoring, group_left
 
     def parse_name_list(self) -> List[str]:
         self.expect("(")
@@ -333,11 +335,23 @@ def _eval(node: Expr, samples: Vector):
             return [Sample(s.name, s.labels, opf(s.value, rhs)) for s in lhs]
         if isinstance(lhs, float):
             return [Sample(s.name, s.labels, opf(lhs, s.value)) for s in rhs]
-        # vector-vector matching
+        # vector-vector matching: key = on-labels, or all labels minus the
+        # ignoring-set, or (default) the full label set
         on = node.on
+        ignoring = node.ignoring
+
+        def match_key(s: Sample) -> Labels:
+            if on is not None:
+                return s.label_key(on)
+            if ignoring is not None:
+                return tuple(sorted(
+                    (k, v) for k, v in s.labels.items()
+                    if k != "__name__" and k not in ignoring))
+            return mklabels(s.labels)
+
         right_index: Dict[Labels, Sample] = {}
         for s in rhs:
-            key = s.label_key(on) if on is not None else mklabels(s.labels)
+            key = match_key(s)
             if key in right_index:
                 raise PromQLError(
                     f"many-to-many matching: duplicate right-side key {key}"
@@ -346,7 +360,7 @@ def _eval(node: Expr, samples: Vector):
         out: Vector = []
         seen_left: Dict[Labels, int] = {}
         for s in lhs:
-            key = s.label_key(on) if on is not None else mklabels(s.labels)
+            key = match_key(s)
             r = right_index.get(key)
             if r is None:
                 continue

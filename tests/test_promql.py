@@ -159,3 +159,33 @@ class TestPrimitives:
             evaluate("m{a=}", [])
         with pytest.raises(PromQLError):
             evaluate("m n", [])
+
+
+class TestIgnoring:
+    def test_ignoring_drops_named_labels_from_key(self):
+        samples = [
+            Sample("l", {"pod": "p", "gpu": "0"}, 3.0),
+            Sample("r", {"pod": "p"}, 2.0),
+        ]
+        # default matching fails (label sets differ); ignoring(gpu) matches
+        assert evaluate("l * r", samples) == []
+        res = evaluate("l * ignoring(gpu) r", samples)
+        assert len(res) == 1 and res[0].value == 6.0
+
+    def test_ignoring_many_to_many_rejected(self):
+        samples = [
+            Sample("l", {"pod": "p", "gpu": "0"}, 1.0),
+            Sample("r", {"pod": "p", "gpu": "0"}, 1.0),
+            Sample("r", {"pod": "p", "gpu": "1"}, 1.0),
+        ]
+        with pytest.raises(PromQLError, match="many-to-many"):
+            evaluate("l * ignoring(gpu) r", samples)
+
+    def test_ignoring_with_group_left(self):
+        samples = [
+            Sample("l", {"pod": "p", "gpu": "0"}, 2.0),
+            Sample("l", {"pod": "p", "gpu": "1"}, 3.0),
+            Sample("r", {"pod": "p"}, 10.0),
+        ]
+        res = evaluate("l * ignoring(gpu) group_left() r", samples)
+        assert sorted(s.value for s in res) == [20.0, 30.0]
