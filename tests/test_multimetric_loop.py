@@ -145,6 +145,10 @@ def test_multimetric_scales_on_bandwidth_gpu(tmp_path):
             )
             metrics = [MetricTarget("cuda_test_gpu_avg", 40.0),
                        MetricTarget("cuda_test_hbm_bw_avg", 20.0)]
+            # the multi-metric HPA keeps its own state (loop.step()'s
+            # internal single-metric reconcile must not share it)
+            from mi355x_gpu_hpa.control import HpaState
+            mm_state = HpaState()
             bw_vals, busy_vals, replicas = [], [], 1
             deadline = time.monotonic() + 10
             while time.monotonic() < deadline:
@@ -155,7 +159,7 @@ def test_multimetric_scales_on_bandwidth_gpu(tmp_path):
                     bw_vals.append(bw)
                     busy_vals.append(busy)
                     replicas = reconcile_multi(
-                        loop.hpa_spec, loop.hpa_state, metrics,
+                        loop.hpa_spec, mm_state, metrics,
                         {"cuda_test_gpu_avg": busy,
                          "cuda_test_hbm_bw_avg": bw}, time.monotonic())
                 time.sleep(0.2)
