@@ -490,3 +490,38 @@ def validate_manifest(doc: dict) -> List[SchemaError]:
     _check(schema, doc, kind, errors)
     _cross_checks(doc, errors)
     return errors
+
+
+def _main(argv):
+    """CLI: python -m mi355x_gpu_hpa.k8s_schema deploy/*.yaml — the offline
+    stand-in for kubectl's server-side validation. Exit 1 on any error."""
+    import sys
+
+    import yaml
+
+    rc = 0
+    for path in argv:
+        try:
+            docs = [d for d in yaml.safe_load_all(open(path)) if d]
+        except Exception as e:  # noqa: BLE001
+            print(f"{path}: unparseable YAML: {e}")
+            rc = 1
+            continue
+        for doc in docs:
+            kind = doc.get("kind", "?")
+            if kind == "Kustomization" or "kind" not in doc:
+                continue  # not a k8s object (kustomize / Helm values)
+            errors = validate_manifest(doc)
+            if errors:
+                rc = 1
+                for e in errors:
+                    print(f"{path} [{kind}]: {e}")
+            else:
+                print(f"{path} [{kind}]: OK")
+    return rc
+
+
+if __name__ == "__main__":
+    import sys
+
+    sys.exit(_main(sys.argv[1:]))
