@@ -129,7 +129,7 @@ class TestReferenceManifestsAgainstModernSchema:
         assert validate_manifest(doc) == []
 
 
-def test_cli_validates_shipped_manifests():
+def test_cli_validates_shipped_manifests(tmp_path):
     """`python -m mi355x_gpu_hpa.k8s_schema deploy/*.yaml` — the operator
     CLI exits 0 on the shipped surface and 1 on a broken manifest."""
     import subprocess
@@ -142,17 +142,12 @@ def test_cli_validates_shipped_manifests():
     assert p.returncode == 0, p.stdout.decode()
     assert p.stdout.decode().count("OK") >= len(files)
 
-    bad = DEPLOY.parent / "tests" / "fixtures"
-    bad.mkdir(exist_ok=True)
-    f = bad / "bad_manifest.yaml"
+    f = tmp_path / "bad_manifest.yaml"
     f.write_text("apiVersion: autoscaling/v2beta1\nkind: HorizontalPodAutoscaler\n"
                  "metadata: {name: x}\nspec: {maxReplicas: 3,\n"
                  "  scaleTargetRef: {kind: Deployment, name: x}}\n")
-    try:
-        p = subprocess.run([sys.executable, "-m", "mi355x_gpu_hpa.k8s_schema",
-                            str(f)], capture_output=True, timeout=60,
-                           cwd=DEPLOY.parent)
-        assert p.returncode == 1
-        assert b"v2beta1" in p.stdout
-    finally:
-        f.unlink()
+    p = subprocess.run([sys.executable, "-m", "mi355x_gpu_hpa.k8s_schema",
+                        str(f)], capture_output=True, timeout=60,
+                       cwd=DEPLOY.parent)
+    assert p.returncode == 1
+    assert b"v2beta1" in p.stdout
