@@ -4,10 +4,12 @@
 Usage: python tools/gemm_ab_gpu.py [variants...] [--shapes 8192 16k8k]
 Runs each variant's bench entry several times interleaved (A,B,A,B,...) so
 DVFS drift hits all variants equally; prints per-run TF/s and the max.
-Variant map: 2=product d6+raster, 7=d7 5-barrier, 8=d6 no-raster,
-10=soft-lgkm d6, 11=d9 single-barrier+raster, 12=d9 no-raster, 13=d14
-16-wave 4-waves/SIMD, 14=d6 round-1 product (now ablation),
-17=d18 3-deep-B rotation (160 KiB LDS), 18=d19 pipelined-reads, 19=d20 4-wave AGPR-acc, 100/101=fp8 E4M3 kernel with/without raster.
+Variant map: 2=PRODUCT (auto d9/d18 by grid size), 7=d7 5-barrier,
+8=d6 no-raster, 10=soft-lgkm d6, 11=d9 single-barrier+raster,
+12=d9 no-raster, 13=d14 16-wave 4-waves/SIMD, 14=d6 round-1 product,
+15=d9 wide-epilogue, 16=d9e burst-staging, 17=d18 3-deep-B (160 KiB LDS),
+18=d19 pipelined reads, 19=d20 4-wave AGPR-acc, 20=d21 d20+deep-B+prefetch,
+100/101/102=fp8 E4M3 raster/no-raster/deep-B.
 """
 
 import json
