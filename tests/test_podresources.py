@@ -37,59 +37,17 @@ def _lib():
     return lib
 
 
-# --- protobuf encoding helpers (mirror the kubelet v1 API messages) --------
+# --- protobuf encoding helpers: the shared kubelet wire encoder ----------
 
-def _tag(field, wire):
-    return bytes([(field << 3) | wire])
-
-
-def _varint(n):
-    out = b""
-    while True:
-        b = n & 0x7F
-        n >>= 7
-        if n:
-            out += bytes([b | 0x80])
-        else:
-            out += bytes([b])
-            return out
-
-
-def _ld(field, payload: bytes):
-    return _tag(field, 2) + _varint(len(payload)) + payload
-
-
-def _s(field, s: str):
-    return _ld(field, s.encode())
-
-
-def container_devices(resource, ids):
-    out = _s(1, resource)
-    for i in ids:
-        out += _s(2, i)
-    return out
-
-
-def container(name, devices):
-    out = _s(1, name)
-    for d in devices:
-        out += _ld(2, d)
-    return out
-
-
-def pod(name, ns, containers):
-    out = _s(1, name) + _s(2, ns)
-    for c in containers:
-        out += _ld(3, c)
-    return out
-
-
-def list_response(pods):
-    out = b""
-    for p in pods:
-        out += _ld(1, p)
-    return out
-
+from mi355x_gpu_hpa.podresources_wire import (  # noqa: E402
+    _ld,
+    _tag,
+    _varint,
+    container,
+    container_devices,
+    list_response,
+    pod,
+)
 
 SAMPLE_RESPONSE = list_response([
     pod("cuda-test-abc", "default", [

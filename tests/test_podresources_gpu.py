@@ -14,7 +14,7 @@ from concurrent import futures
 
 import pytest
 
-from tests.test_podresources import (  # noqa: F401  (reuse proto encoders)
+from mi355x_gpu_hpa.podresources_wire import (
     container,
     container_devices,
     list_response,

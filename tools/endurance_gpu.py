@@ -35,8 +35,7 @@ from mi355x_gpu_hpa.control import (  # noqa: E402
 )
 from mi355x_gpu_hpa.exporter import ExporterProcess  # noqa: E402
 
-sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
-from test_podresources import (  # noqa: E402
+from mi355x_gpu_hpa.podresources_wire import (  # noqa: E402
     container,
     container_devices,
     list_response,
