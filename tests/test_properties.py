@@ -122,3 +122,76 @@ class TestNativeRendererProperties:
                 assert math.isclose(s.value, busy, rel_tol=1e-5, abs_tol=1e-4)
         finally:
             del os.environ["MI355X_MOCK_BUSY"]
+
+
+class TestSchemaValidatorRobustness:
+    """validate_manifest must never crash: any JSON-ish document yields a
+    (possibly long) error list, not an exception."""
+
+    json_scalars = st.one_of(st.none(), st.booleans(),
+                             st.integers(-10**6, 10**6),
+                             st.floats(allow_nan=False, allow_infinity=False),
+                             st.text(max_size=20))
+    json_values = st.recursive(
+        json_scalars,
+        lambda children: st.one_of(
+            st.lists(children, max_size=4),
+            st.dictionaries(st.text(max_size=12), children, max_size=4)),
+        max_leaves=25)
+
+    @given(st.dictionaries(st.text(max_size=12), json_values, max_size=6))
+    @settings(max_examples=300, deadline=None)
+    def test_never_raises_on_random_docs(self, doc):
+        from mi355x_gpu_hpa.k8s_schema import validate_manifest
+
+        errors = validate_manifest(doc)
+        assert isinstance(errors, list)
+
+    @given(st.sampled_from(["DaemonSet", "Deployment", "Service",
+                            "HorizontalPodAutoscaler", "PrometheusRule",
+                            "ConfigMap"]),
+           st.dictionaries(st.text(max_size=12), json_values, max_size=5))
+    @settings(max_examples=300, deadline=None)
+    def test_never_raises_on_random_typed_docs(self, kind, extra):
+        from mi355x_gpu_hpa.k8s_schema import validate_manifest
+
+        doc = dict(extra)
+        doc["kind"] = kind
+        errors = validate_manifest(doc)
+        assert isinstance(errors, list)
+        assert errors  # a random doc of a pinned kind cannot be valid
+
+
+class TestAdapterDiscoveryProperties:
+    labels = st.dictionaries(
+        st.sampled_from(["namespace", "pod", "deployment", "node", "zzz",
+                         "service", "job", "gpu"]),
+        st.text(alphabet="abcdefg-", min_size=0, max_size=8), max_size=5)
+    samples = st.lists(
+        st.builds(lambda n, l, v: __import__(
+            "mi355x_gpu_hpa.control", fromlist=["Sample"]).Sample(n, l, v),
+            st.sampled_from(["m1", "m2_total", "container_x", "",
+                             "dcgm_gpu_utilization"]),
+            labels,
+            st.floats(allow_nan=False, allow_infinity=False)),
+        max_size=12)
+
+    @given(samples)
+    @settings(max_examples=300, deadline=None)
+    def test_discovery_invariants(self, samples):
+        from mi355x_gpu_hpa.control import discover
+
+        d = discover(samples)
+        for metric, m in d.items():
+            # only namespace-labeled, non-container series are discovered
+            assert not m.series.startswith("container_")
+            assert any(s.name == m.series and s.labels.get("namespace")
+                       for s in samples)
+            # name mangling strips _total exactly when is_counter
+            assert m.is_counter == m.series.endswith("_total")
+            if m.is_counter:
+                assert not metric.endswith("_total")
+            # resource bindings only from non-empty known labels
+            for r in m.resources:
+                assert any(s.labels.get(r) for s in samples
+                           if s.name == m.series)
