@@ -252,6 +252,7 @@ def main():
     # xGMI all-reduce per step when world > 1 ------------------------------
     latencies = []
     busy_seen = []          # the metric value each cycle decided on
+    slowest = None          # (total_s, LoopResult) — p99 attribution
     last_tick = exporter_tick() if rank == 0 else None
     t0 = time.monotonic()
     for _ in range(args.steps):
@@ -260,6 +261,8 @@ def main():
             last_tick = wait_fresh_tick(last_tick)
             r = loop.step()
             latencies.append(r.total_s)
+            if slowest is None or r.total_s > slowest[0]:
+                slowest = (r.total_s, r)
             if r.metric_value is not None:
                 busy_seen.append(r.metric_value)
         if w:
@@ -405,6 +408,15 @@ def main():
                 "load_step_detection_s": (round(load_step_detection_s, 4)
                                           if load_step_detection_s is not None
                                           else None),
+                # where the slowest cycle's time went (p99 attribution:
+                # scrape = exporter HTTP GET + parse; the rest is in-process)
+                "slowest_cycle_ms": ({
+                    "total": round(slowest[0] * 1e3, 3),
+                    "scrape": round(slowest[1].scrape_s * 1e3, 3),
+                    "rule": round(slowest[1].rule_eval_s * 1e3, 3),
+                    "adapter": round(slowest[1].adapter_s * 1e3, 3),
+                    "hpa": round(slowest[1].hpa_s * 1e3, 3),
+                } if slowest else None),
             },
         }
         print(json.dumps(result), flush=True)
