@@ -48,6 +48,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=30.0)
     ap.add_argument("--interval-ms", type=float, default=100.0)
+    ap.add_argument("--backend", default=None)
     args = ap.parse_args()
 
     stop = ctypes.c_int(0)
@@ -62,7 +63,8 @@ def main():
     intervals = []
     scrape_ms = []
     try:
-        with ExporterProcess(interval_ms=args.interval_ms) as exp:
+        with ExporterProcess(interval_ms=args.interval_ms,
+                         backend=args.backend) as exp:
             last_acc = None
             last_change = None
             t_end = time.monotonic() + args.seconds
