@@ -47,12 +47,21 @@ int main(int argc, char** argv)
     if (cfg.mock_devices > 0) {
         backend = make_mock_backend(cfg.mock_devices, cfg.mock_busy_file);
     } else {
-        // auto: prefer amd-smi (rocm_smi_lib is in maintenance mode
-        // upstream), fall back to rocm_smi; --backend pins one.
+        // auto selection is CADENCE-AWARE: libamd_smi caches gpu_metrics
+        // internally for ~100-200 ms, so at sub-250 ms ticks every other
+        // sample would repeat stale accumulator values (measured:
+        // update-interval p90 206 ms via amd-smi vs 102 ms via rocm_smi
+        // at a 100 ms tick — profiles/exporter_cadence_jitter.md). Fast
+        // cadences therefore prefer rocm_smi; the DaemonSet-default 1 s
+        // tick prefers amd-smi (rocm_smi_lib is in maintenance mode
+        // upstream). --backend pins one explicitly.
         std::string err_amdsmi, err_rsmi;
-        if (cfg.backend == "auto" || cfg.backend == "amdsmi")
+        bool fast_tick = cfg.interval_ms < 250.0;
+        if (cfg.backend == "rsmi" || (cfg.backend == "auto" && fast_tick))
+            backend = make_rsmi_backend(&err_rsmi);
+        if (!backend && (cfg.backend == "auto" || cfg.backend == "amdsmi"))
             backend = make_amdsmi_backend(&err_amdsmi);
-        if (!backend && (cfg.backend == "auto" || cfg.backend == "rsmi"))
+        if (!backend && cfg.backend == "auto" && !fast_tick)
             backend = make_rsmi_backend(&err_rsmi);
         if (!backend) {
             std::fprintf(stderr,
