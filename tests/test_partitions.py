@@ -187,3 +187,23 @@ class TestExporterObservability:
                            timeout=10)
         assert p.returncode == 2
         assert b"bad backend" in p.stderr
+
+
+@pytest.mark.gpu
+class TestCadenceAwareAutoSelection:
+    """`--backend auto` must pick rocm_smi below 250 ms (libamd_smi's
+    internal gpu_metrics cache would serve stale accumulators there —
+    profiles/exporter_cadence_jitter.md) and amd-smi at the DaemonSet
+    default cadence."""
+
+    @pytest.mark.parametrize("interval_ms,expected", [
+        (100, b"backend=rocm_smi"),
+        (1000, b"backend=amd_smi"),
+    ])
+    def test_auto_backend_by_cadence(self, gpu, interval_ms, expected):
+        with ExporterProcess(interval_ms=interval_ms) as exp:
+            time.sleep(0.3)
+            proc = exp.proc
+            exp.terminate()
+            err = proc.stderr.read()
+        assert expected in err, err[:300]
