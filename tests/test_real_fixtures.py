@@ -69,3 +69,42 @@ def test_reference_rule_on_real_series(samples):
     assert v is not None
     util = [x.value for x in s if x.name == "dcgm_gpu_utilization"][0]
     assert v == util
+
+
+class TestRound2Fixture:
+    """Fresh round-2 capture (auto backend, full surface incl. partitions,
+    probes and the freshness counter) — validates the new families'
+    counter-to-metric math against genuine gfx950 output, CPU-only."""
+
+    @pytest.fixture()
+    def r2(self):
+        text = (FIXTURES / "real_mi355x_idle_r2.prom").read_text()
+        return parse_prometheus_text(text)
+
+    def test_partition_families_on_real_hw(self, r2):
+        info = [s for s in r2 if s.name == "amd_compute_partition_info"]
+        assert info and info[0].labels["compute"] == "SPX"
+        assert info[0].labels["memory"].startswith("NPS")
+        xcp = [s for s in r2 if s.name == "amd_xcp_busy_percent"]
+        assert xcp and all(0 <= s.value <= 100 for s in xcp)
+        assert xcp[0].labels["partition"] == "0"  # SPX = one partition
+
+    def test_probe_family_on_real_hw(self, r2):
+        un = {s.labels["counter"]: s.labels["reason"] for s in r2
+              if s.name == "amd_counter_unavailable"}
+        assert "mfma_activity" in un  # probed, honestly absent
+
+    def test_freshness_and_overhead_families(self, r2):
+        ticks = [s for s in r2 if s.name == "amd_exporter_samples_total"]
+        assert ticks and ticks[0].value >= 1
+        dur = [s for s in r2 if s.name == "amd_exporter_sample_duration_ms"]
+        assert dur and 0 <= dur[0].value < 100
+
+    def test_reference_rule_on_real_output(self, r2):
+        for s in r2:
+            if s.labels.get("gpu") == "0" and "pod" not in s.labels:
+                s.labels["pod"] = "cuda-test-0"
+                s.labels.setdefault("namespace", "default")
+        v = evaluate_scalar(REFERENCE_RULE_EXPR,
+                            r2 + synth_pod_labels(["cuda-test-0"]))
+        assert v is not None and 0 <= v <= 100
