@@ -125,9 +125,10 @@ def test_live_schema_superset_of_fixtures(gpu):
     exporter no longer serves (fixture drift guard)."""
     from pathlib import Path
 
-    fixture = (Path(__file__).parent / "fixtures" /
-               "real_mi355x_idle.prom").read_text()
-    fixture_families = {s.name for s in parse_prometheus_text(fixture)}
+    fixture_families = set()
+    for name in ("real_mi355x_idle.prom", "real_mi355x_idle_r2.prom"):
+        text = (Path(__file__).parent / "fixtures" / name).read_text()
+        fixture_families |= {s.name for s in parse_prometheus_text(text)}
     with ExporterProcess(interval_ms=200) as exp:
         time.sleep(0.5)
         live_families = {s.name for s in parse_prometheus_text(exp.scrape())}
