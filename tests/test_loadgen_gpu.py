@@ -168,13 +168,27 @@ def test_vector_add_loop_runs(gpu):
     assert ms > 0
 
 
+def _floor_check(bench_fn, floor_tf, what):
+    """Fallback detector, not a perf test: an implausibly low reading is
+    re-measured once after a settle (observed one 2.5 TF/s transient when
+    this ran right after a 20 s HBM burn on one box; isolated re-runs
+    measured 450 TF/s — gpurun_out/gputest_final3.log)."""
+    import time as _t
+
+    ms, tf = bench_fn()
+    if tf <= floor_tf:
+        _t.sleep(2.0)
+        ms, tf = bench_fn()
+    assert ms > 0
+    assert tf > floor_tf, f"{what} at {tf:.0f} TF/s — MFMA path not engaged?"
+
+
 def test_gemm_bench_sane(gpu):
     lg = _loadgen()
-    ms, tf = lg.gemm_bench(m=2048, n=2048, k=2048, warmup=2, iters=10)
-    assert ms > 0
     # MFMA path must be in play: even an untuned MFMA GEMM clears 100 TF/s;
     # a VALU/eager fallback cannot.
-    assert tf > 100, f"bf16 GEMM at {tf:.0f} TF/s — MFMA path not engaged?"
+    _floor_check(lambda: lg.gemm_bench(m=2048, n=2048, k=2048, warmup=2,
+                                       iters=10), 100, "bf16 GEMM")
 
 
 def test_bw_burn_hits_hbm(gpu):
@@ -213,11 +227,10 @@ def test_gemm_fp8_numerics(gpu, m, n, k, raster):
 
 def test_gemm_fp8_bench_sane(gpu):
     """The fp8 path must engage the 2x-peak matrix pipe: even untuned,
-    >200 TF/s at a small shape; a non-MFMA path cannot."""
+    >200 TF/s at a small shape (measured 450); a non-MFMA path cannot."""
     lg = _loadgen()
-    ms, tf = lg.gemm_fp8_bench(m=2048, n=2048, k=2048, warmup=2, iters=10)
-    assert ms > 0
-    assert tf > 200, f"fp8 GEMM at {tf:.0f} TF/s — MFMA fp8 path not engaged?"
+    _floor_check(lambda: lg.gemm_fp8_bench(m=2048, n=2048, k=2048, warmup=2,
+                                           iters=10), 200, "fp8 GEMM")
 
 
 def test_gemm_fp8_burn_tracks_target(gpu):
