@@ -22,7 +22,7 @@ SKIP = {"kustomization.yaml", "kube-prometheus-stack-values.yaml"}
 
 MANIFESTS = sorted(
     p for p in list(DEPLOY.glob("*.yaml")) + list(DEPLOY.glob("*/*.yaml"))
-    if p.name not in SKIP and p.parent.name != "kind"
+    if p.name not in SKIP
 )
 
 
