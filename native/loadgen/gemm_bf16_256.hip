@@ -16,8 +16,15 @@
 //   staging         __builtin_amdgcn_global_load_lds 16 B/lane; one
 //                   half-tile = 2 glds per wave (16 KiB / (8 waves x 1 KiB))
 //
-// PRODUCT phase schedule (DEPTH 6, measured best — profiles/): phase q
-// computes m-frags {2q, 2q+1} x ALL FOUR n-frags (16 MFMA):
+// PRODUCT (round 2): the d9/d18 single-barrier schedules defined further
+// down (impl9/impl18; auto-selected by grid size in loadgen_lib.cpp —
+// ~1.0 PF/s @8192^3, 1.25 PF/s @16384^3, profiles/gemm_bf16_256_ladder.md).
+// The remainder of this header documents the round-1 d6 schedule, kept as
+// the A/B baseline (variant 14) together with the full measured ablation
+// family (d1..d21).
+//
+// d6 phase schedule: phase q computes m-frags {2q, 2q+1} x ALL FOUR
+// n-frags (16 MFMA):
 //   q0 reads: A frags 0-1 (4x ds_read_b128) + all B frags (8x), B held to q3
 //   q1-q3 reads: 4x A each  -> 16 LDS reads per K-tile (vs 40 for the
 //   quadrant schedules), and every B slot is dead after q0, which lets B
